@@ -1,0 +1,73 @@
+// mpi4torch_amd — MI355X-native autodiff-transparent collectives for PyTorch.
+//
+// Shared declarations: reduction-op enum, dtype maps, error helpers, config.
+//
+// Design (vs the reference helmholtz-analytics/mpi4torch, csrc/extension.cpp):
+// the reference speaks raw MPI from one C++ TU; we are MI355X-first:
+//   * GPU tensors ride RCCL over xGMI on dedicated HIP streams
+//     (csrc/transport.cpp), with hand-written CDNA4 pack/unpack kernels
+//     (csrc/kernels.hip) replacing MPI derived datatypes
+//     (reference csrc/extension.cpp:556-577).
+//   * CPU tensors ride a c10d gloo backend (torchrun bootstrap, no mpirun).
+// Parity targets are cited per-site as reference file:line.
+#pragma once
+
+#include <torch/types.h>
+#include <c10/util/Exception.h>
+
+#include <cstdint>
+#include <string>
+
+namespace m4a {
+
+// Reduction op constants. Mirrors the 12 ops of the reference
+// (csrc/extension.cpp:204-252, bound at :1424-1435). Values are our own
+// stable ABI (the reference exported raw MPI_Op handles cast to int).
+enum RedOp : int64_t {
+  kMax = 0,
+  kMin = 1,
+  kSum = 2,
+  kProd = 3,
+  kLAnd = 4,
+  kBAnd = 5,
+  kLOr = 6,
+  kBOr = 7,
+  kLXor = 8,
+  kBXor = 9,
+  kMinLoc = 10,
+  kMaxLoc = 11,
+};
+
+inline const char* red_op_name(int64_t op) {
+  switch (op) {
+    case kMax: return "MPI_MAX";
+    case kMin: return "MPI_MIN";
+    case kSum: return "MPI_SUM";
+    case kProd: return "MPI_PROD";
+    case kLAnd: return "MPI_LAND";
+    case kBAnd: return "MPI_BAND";
+    case kLOr: return "MPI_LOR";
+    case kBOr: return "MPI_BOR";
+    case kLXor: return "MPI_LXOR";
+    case kBXor: return "MPI_BXOR";
+    case kMinLoc: return "MPI_MINLOC";
+    case kMaxLoc: return "MPI_MAXLOC";
+    default: return "<invalid op>";
+  }
+}
+
+// Global config toggles (see python layer mpi4torch_amd/utils/config.py).
+struct Config {
+  // Mirrors reference deactivate_cuda_aware_mpi_support()
+  // (csrc/extension.cpp:54-59,1404-1414): when true, GPU tensors are staged
+  // through host memory and communicated over the CPU (gloo) transport —
+  // a debugging escape hatch, never the default on MI355X.
+  bool force_host_staging = false;
+  // hipEvent-based per-collective tracing (the reference has
+  // none; we add it as a first-class aux subsystem).
+  bool trace_enabled = false;
+};
+
+Config& config();
+
+} // namespace m4a

@@ -1,0 +1,45 @@
+// Host-side declarations for the CDNA4 (gfx950) HIP kernels in kernels.hip.
+// This header is torch-free so the device TU stays minimal.
+#pragma once
+
+#include <hip/hip_runtime_api.h>
+#include <cstdint>
+
+namespace m4a {
+
+// Describes one slab of a batched strided copy: the MI355X-native
+// replacement for the reference's MPI derived-datatype marshaling
+// (MPI_Type_vector + MPI_Type_create_resized, reference
+// csrc/extension.cpp:556-577, 691-712, 839-861).
+//
+// Copies, for b in [0,before), c in [0,count):
+//   dst[b*dst_pitch_b + c*after_b + 0..after_b) =
+//   src[b*src_pitch_b + c*after_b + 0..after_b)
+// All quantities in BYTES except `before`/`count` (row counts).
+struct SlabDesc {
+  const void* src;
+  void* dst;
+  int64_t before;
+  int64_t count;
+  int64_t after_b;
+  int64_t src_pitch_b;  // byte stride between consecutive `b` on src
+  int64_t dst_pitch_b;  // byte stride between consecutive `b` on dst
+};
+
+// Max slabs handled by a single kernel launch (descriptor array is passed
+// by value in kernel args). Larger batches loop over launches.
+constexpr int kMaxSlabsPerLaunch = 8;
+
+// Batched strided copy of `n` slabs on `stream`. Picks the widest access
+// (16B/4B/1B) every slab admits; one launch per <=8 slabs.
+void launch_slab_copy(const SlabDesc* descs, int n, hipStream_t stream);
+
+// Elementwise bitwise reduction across `nranks` contiguous chunks:
+//   out[0..chunk_bytes) = op_{r<nranks} in[r*chunk_bytes ..]
+// op: 0=AND 1=OR 2=XOR. Bitwise ops are byte-local, so this is
+// dtype-independent (serves MPI_BAND/BOR/BXOR for every integer dtype;
+// reference op table csrc/extension.cpp:204-252).
+void launch_bitwise_reduce(const void* in, void* out, int64_t chunk_bytes,
+                           int nranks, int op, hipStream_t stream);
+
+} // namespace m4a

@@ -221,10 +221,10 @@ bool native_reduce_dtype(const Transport& tr, at::ScalarType t) {
         return false;  // fp8/short/bool go through the upcast path
     }
   }
+  // gloo (like RCCL) has no int16: kShort takes the int32 upcast path
   switch (t) {
     case at::kByte:
     case at::kChar:
-    case at::kShort:
     case at::kInt:
     case at::kLong:
     case at::kHalf:

@@ -1,0 +1,145 @@
+"""Allreduce: forward values, closed-form adjoints, op/dtype coverage.
+
+Mirrors the assertions of reference tests/test_collectives.py:8-21 and
+extends them with the full reduction-op table and the dtype lowering paths
+(bf16 / int16 / fp8) the reference never supported (its dtype map,
+csrc/extension.cpp:106-129, lacked them).
+"""
+
+import pytest
+import torch
+
+from spmd import run_spmd
+
+
+def _sum_worker(rank, world):
+    import mpi4torch_amd as m
+
+    comm = m.COMM_WORLD
+    assert comm.rank == rank
+    assert comm.size == world
+
+    # closed-form adjoint: d(sum-allreduce)/dx = allreduce(ones) = world*ones
+    # (reference tests/test_collectives.py:8-12)
+    tmp = torch.rand(10, dtype=torch.double).requires_grad_()
+    res = comm.Allreduce(tmp, m.MPI_SUM)
+    res.sum().backward()
+    assert (tmp.grad == world * torch.ones(10, dtype=torch.double)).all()
+
+    # forward value
+    t = torch.full((7,), float(rank + 1))
+    assert (comm.Allreduce(t, m.MPI_SUM) == world * (world + 1) / 2).all()
+
+    # non-contiguous input
+    nc = torch.arange(12, dtype=torch.float).reshape(3, 4).t()
+    r = comm.Allreduce(nc, m.MPI_SUM)
+    assert (r == world * nc).all()
+
+
+def _ops_worker(rank, world):
+    import mpi4torch_amd as m
+
+    comm = m.COMM_WORLD
+    v = torch.tensor([rank + 1, world - rank], dtype=torch.float)
+    assert (comm.Allreduce(v, m.MPI_MAX) == torch.tensor([world, world]).float()).all()
+    assert (comm.Allreduce(v, m.MPI_MIN) == torch.tensor([1, 1]).float()).all()
+    import math
+
+    prod = comm.Allreduce(v, m.MPI_PROD)
+    assert prod[0].item() == pytest.approx(math.factorial(world))
+
+    # logical ops (lowered to indicator min/max/sum internally)
+    b = torch.tensor([1, rank % 2, 0], dtype=torch.int32)
+    land = comm.Allreduce(b, m.MPI_LAND)
+    lor = comm.Allreduce(b, m.MPI_LOR)
+    lxor = comm.Allreduce(b, m.MPI_LXOR)
+    n_odd = sum(1 for r in range(world) if r % 2)
+    assert land.tolist() == [1, 1 if n_odd == world else 0, 0]
+    assert lor.tolist() == [1, 1 if n_odd > 0 else 0, 0]
+    assert lxor.tolist() == [world % 2, n_odd % 2, 0]
+
+    # bitwise ops
+    x = torch.tensor([0b1100 | rank, 0b1010], dtype=torch.int64)
+    band = comm.Allreduce(x, m.MPI_BAND)
+    bor = comm.Allreduce(x, m.MPI_BOR)
+    exp_and = 0b1100 | rank
+    exp_or = 0b1100 | rank
+    for r in range(world):
+        exp_and &= 0b1100 | r
+        exp_or |= 0b1100 | r
+    assert band[0].item() == exp_and and band[1].item() == 0b1010
+    assert bor[0].item() == exp_or and bor[1].item() == 0b1010
+    bxor = comm.Allreduce(torch.tensor([rank], dtype=torch.int32), m.MPI_BXOR)
+    exp_xor = 0
+    for r in range(world):
+        exp_xor ^= r
+    assert bxor.item() == exp_xor
+
+
+def _dtypes_worker(rank, world):
+    import mpi4torch_amd as m
+
+    comm = m.COMM_WORLD
+    for dtype in (
+        torch.float64,
+        torch.float32,
+        torch.float16,
+        torch.bfloat16,
+        torch.int64,
+        torch.int32,
+        torch.int16,
+        torch.uint8,
+    ):
+        t = torch.ones(5, dtype=dtype)
+        r = comm.Allreduce(t, m.MPI_SUM)
+        assert r.dtype == dtype
+        assert (r.float() == world).all(), (dtype, r)
+    # fp8: upcast-reduce-downcast path (reference had no fp8 at all)
+    t8 = torch.ones(4, dtype=torch.float8_e4m3fn)
+    r8 = comm.Allreduce(t8, m.MPI_SUM)
+    assert r8.dtype == torch.float8_e4m3fn
+    assert (r8.float() == world).all()
+
+
+def _errors_worker(rank, world):
+    import mpi4torch_amd as m
+
+    comm = m.COMM_WORLD
+    t = torch.rand(4).requires_grad_()
+    # non-SUM backward is unimplemented, matching the reference (N7)
+    res = comm.Allreduce(t, m.MPI_MAX)
+    try:
+        res.sum().backward()
+        raise AssertionError("expected RuntimeError for MAX backward")
+    except RuntimeError:
+        pass
+    # MINLOC/MAXLOC have no tensor equivalent
+    try:
+        comm.Allreduce(torch.rand(4), m.MPI_MINLOC)
+        raise AssertionError("expected RuntimeError for MINLOC")
+    except RuntimeError:
+        pass
+
+
+def test_allreduce_sum_ws2():
+    run_spmd(2, _sum_worker)
+
+
+def test_allreduce_sum_ws5():
+    run_spmd(5, _sum_worker)
+
+
+def test_allreduce_ops_ws2():
+    run_spmd(2, _ops_worker)
+
+
+def test_allreduce_ops_ws7():
+    run_spmd(7, _ops_worker)
+
+
+def test_allreduce_dtypes_ws2():
+    run_spmd(2, _dtypes_worker)
+
+
+def test_allreduce_errors_ws2():
+    run_spmd(2, _errors_worker)

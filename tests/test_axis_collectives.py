@@ -1,0 +1,212 @@
+"""Gather/Allgather/Scatter/Alltoall: axis semantics, round-trip identities,
+closed-form adjoints, variable per-rank counts.
+
+Mirrors reference tests/test_collectives.py:48-147 (including the algebraic
+identities Scatter∘Gather = id, Alltoall ≡ Scatter∘Gather,
+Alltoall∘Alltoall = id, and the same-axis variable-count repartition), plus
+the corrected Allgather adjoint (reduce-scatter; the reference's composite
+had the wrong-root bug at csrc/extension.cpp:626-628 that uniform-gradient
+tests cannot catch — we add a NON-uniform gradient test that does).
+"""
+
+import torch
+
+from spmd import run_spmd
+
+
+def _gather_worker(rank, world):
+    import mpi4torch_amd as m
+
+    comm = m.COMM_WORLD
+    numdim = 4
+    # middle-axis gather exercises the strided pack path (before > 1)
+    tmp = torch.rand([2, 5, numdim, 2, 3], dtype=torch.double)
+    tmp[0, 0, :, 0, 0] = comm.rank
+    res = comm.Gather(tmp, 2, 0)
+    if rank == 0:
+        s = torch.sum(res[0, 0, :, 0, 0])
+        assert s == numdim * (world - 1) * world // 2
+        assert list(res.shape) == [2, 5, numdim * world, 2, 3]
+        # verify interleaving: block r occupies [r*numdim, (r+1)*numdim)
+        for r in range(world):
+            assert (res[0, 0, r * numdim : (r + 1) * numdim, 0, 0] == r).all()
+    else:
+        assert res.shape[2] == 0
+
+    # adjoint: ones flow back to every rank (reference :58-63)
+    t2 = torch.rand([2, 5, numdim, 2, 3], dtype=torch.double).requires_grad_()
+    comm.Gather(t2, 2, 0).sum().backward()
+    assert (t2.grad == torch.ones_like(t2)).all()
+
+    # variable axis sizes
+    tv = torch.full((3, rank + 1, 2), float(rank), dtype=torch.double)
+    rv = comm.Gather(tv, 1, 0)
+    if rank == 0:
+        total = world * (world + 1) // 2
+        assert list(rv.shape) == [3, total, 2]
+        off = 0
+        for r in range(world):
+            assert (rv[:, off : off + r + 1, :] == r).all()
+            off += r + 1
+
+
+def _allgather_worker(rank, world):
+    import mpi4torch_amd as m
+
+    comm = m.COMM_WORLD
+    numdim = 4
+    tmp = torch.rand([2, 5, numdim, 2, 3], dtype=torch.double)
+    tmp[0, 0, :, 0, 0] = comm.rank
+    res = comm.Allgather(tmp, 2)
+    s = torch.sum(res[0, 0, :, 0, 0])
+    assert s == numdim * (world - 1) * world // 2
+
+    # uniform-grad adjoint (reference :77-82)
+    t2 = torch.rand([2, 5, numdim, 2, 3], dtype=torch.double).requires_grad_()
+    comm.Allgather(t2, 2).sum().backward()
+    assert (t2.grad == world * torch.ones_like(t2)).all()
+
+    # NON-uniform gradient: catches the reference's wrong-root adjoint bug.
+    # grad of output slice r is (r+1); correct adjoint (reduce-scatter)
+    # gives every rank grad = world * (rank+1) on its own slice.
+    t3 = torch.rand([3, numdim, 2], dtype=torch.double).requires_grad_()
+    res3 = comm.Allgather(t3, 1)
+    weight = torch.zeros_like(res3)
+    for r in range(world):
+        weight[:, r * numdim : (r + 1) * numdim, :] = r + 1
+    (res3 * weight).sum().backward()
+    assert (t3.grad == world * (rank + 1) * torch.ones_like(t3)).all()
+
+    # variable counts
+    tv = torch.full((2, rank + 1), float(rank), dtype=torch.double)
+    rv = comm.Allgather(tv, 1)
+    total = world * (world + 1) // 2
+    assert list(rv.shape) == [2, total]
+    off = 0
+    for r in range(world):
+        assert (rv[:, off : off + r + 1] == r).all()
+        off += r + 1
+
+
+def _scatter_worker(rank, world):
+    import mpi4torch_amd as m
+
+    comm = m.COMM_WORLD
+    # reference tests/test_collectives.py:85-113
+    if rank == 0:
+        tmp = torch.rand([2, 5, world, 2, 3], dtype=torch.double)
+        for i in range(world):
+            tmp[0, 0, i, 0, 0] = i
+    else:
+        tmp = torch.rand([1], dtype=torch.double)
+    res = comm.Scatter(tmp, 2, 1, 0)
+    assert (res[0, 0, :, 0, 0] == rank).all()
+    assert list(res.shape) == [2, 5, 1, 2, 3]
+
+    # Scatter∘Gather = identity
+    res2 = comm.Gather(res, 2, 0)
+    if rank == 0:
+        assert (res2 == tmp).all()
+
+    # adjoint: root gets ones, non-root zeros
+    if rank == 0:
+        t = torch.rand([2, 5, world, 2, 3], dtype=torch.double).requires_grad_()
+    else:
+        t = torch.rand([1], dtype=torch.double).requires_grad_()
+    comm.Scatter(t, 2, 1, 0).sum().backward()
+    if rank == 0:
+        assert (t.grad == torch.ones_like(t)).all()
+    else:
+        assert (t.grad == torch.zeros_like(t)).all()
+
+    # variable counts from a non-zero root
+    root = world - 1
+    if rank == root:
+        total = world * (world + 1) // 2
+        tv = torch.empty([total, 3], dtype=torch.double)
+        off = 0
+        for r in range(world):
+            tv[off : off + r + 1, :] = r
+            off += r + 1
+    else:
+        tv = torch.rand([1], dtype=torch.double)
+    rv = comm.Scatter(tv, 0, rank + 1, root)
+    assert list(rv.shape) == [rank + 1, 3]
+    assert (rv == rank).all()
+
+
+def _alltoall_worker(rank, world):
+    import mpi4torch_amd as m
+
+    comm = m.COMM_WORLD
+    # Alltoall ≡ Scatter∘Gather (reference :115-125)
+    tmp = torch.rand([3, 4, 1, 4, world, 2], dtype=torch.double)
+    res1 = comm.Scatter(comm.Gather(tmp, 2, 0), 4, 1, 0)
+    res2 = comm.Alltoall(tmp, 2, 4, 1)
+    assert (res2 == res1).all()
+
+    # with varying numelem (reference :121-125)
+    tot = world * (world + 1) // 2
+    tmp = torch.rand([3, 4, rank + 1, 4, tot, 2], dtype=torch.double)
+    res1 = comm.Scatter(comm.Gather(tmp, 2, 0), 4, rank + 1, 0)
+    res2 = comm.Alltoall(tmp, 2, 4, rank + 1)
+    assert (res2 == res1).all()
+
+    # same-axis repartition with variable counts (reference :127-135)
+    tmp = torch.rand([3, 4, rank + 1, 2], dtype=torch.double)
+    tmp[0, 0, :, 0] = torch.arange(
+        rank * (rank + 1) // 2, (rank + 1) * (rank + 2) // 2
+    )
+    res = comm.Alltoall(tmp, 2, 2, world - rank)
+    total = world * (world + 1) // 2
+    lo = total - (world - rank) * (world - rank + 1) // 2
+    hi = total - (world - rank - 1) * (world - rank) // 2
+    correct = torch.arange(lo, hi, dtype=torch.double)
+    assert (res[0, 0, :, 0] == correct).all()
+
+    # Alltoall∘Alltoall = identity (reference :137-141)
+    tmp = torch.rand([3, 4, 2, 4, 3 * world, 2], dtype=torch.double)
+    r1 = comm.Alltoall(tmp, 2, 4, 3)
+    r2 = comm.Alltoall(r1, 4, 2, 2)
+    assert (r2 == tmp).all()
+
+    # adjoint = ones (reference :143-147)
+    t = torch.rand([3, 4, 2, 4, world, 2], dtype=torch.double).requires_grad_()
+    comm.Alltoall(t, 2, 4, 1).sum().backward()
+    assert (t.grad == torch.ones_like(t)).all()
+
+
+def test_gather_ws2():
+    run_spmd(2, _gather_worker)
+
+
+def test_gather_ws5():
+    run_spmd(5, _gather_worker)
+
+
+def test_allgather_ws2():
+    run_spmd(2, _allgather_worker)
+
+
+def test_allgather_ws5():
+    run_spmd(5, _allgather_worker)
+
+
+def test_scatter_ws2():
+    run_spmd(2, _scatter_worker)
+
+
+def test_scatter_ws5():
+    run_spmd(5, _scatter_worker)
+
+
+def test_alltoall_ws2():
+    run_spmd(2, _alltoall_worker)
+
+
+def test_alltoall_ws5():
+    run_spmd(5, _alltoall_worker)
+
+
+def test_alltoall_ws7():
+    run_spmd(7, _alltoall_worker)

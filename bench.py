@@ -1,0 +1,255 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: autograd-transparent Allreduce fwd+bwd on a 1 GiB
+bf16 tensor per GPU (BASELINE.json config #2 — the headline metric:
+"Allreduce algbw GB/s + grad-step wallclock, 1GiB bf16 tensor").
+
+One step = forward Allreduce(SUM) of the 1 GiB tensor + backward through
+its autograd node (the adjoint Allreduce of a 1 GiB gradient) — i.e. the
+full autodiff-transparent round trip, 2 GiB of collective payload per GPU
+per step. The reported value is the WHOLE-JOB algorithmic bandwidth:
+  value = n_gpus * 2 * bytes / t_step      [GB/s, higher is better]
+`vs_baseline` is null: the reference publishes no numbers.
+
+Launch (driver contract):
+  python bench.py --gpus 1 --steps K --warmup W          # single GPU
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N ...      # N ranks, RCCL
+
+Other BASELINE configs are reachable with --config {allreduce,linreg,
+ring,alltoall} for self-measurement; the driver uses the default.
+"""
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def setup_world(args):
+    if "WORLD_SIZE" not in os.environ:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = os.environ.get("MASTER_PORT", "29400")
+        os.environ["RANK"] = "0"
+        os.environ["WORLD_SIZE"] = "1"
+        os.environ["LOCAL_RANK"] = "0"
+    import torch.distributed as dist
+
+    rank = int(os.environ["RANK"])
+    world = int(os.environ["WORLD_SIZE"])
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    if not dist.is_initialized():
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+    if torch.cuda.is_available():
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
+    else:
+        device = torch.device("cpu")
+    import mpi4torch_amd as m4a
+
+    m4a.init()
+    return m4a, rank, world, device
+
+
+def sync(device):
+    if device.type == "cuda":
+        torch.cuda.synchronize(device)
+
+
+def time_steps(step_fn, steps, warmup, device):
+    import torch.distributed as dist
+
+    for _ in range(warmup):
+        step_fn()
+    sync(device)
+    dist.barrier()
+    sync(device)
+    t0 = time.perf_counter()
+    for _ in range(steps):
+        step_fn()
+    sync(device)
+    elapsed = time.perf_counter() - t0
+    dist.barrier()
+    # MAX over ranks
+    t = torch.tensor([elapsed], dtype=torch.float64)
+    dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    return float(t.item())
+
+
+def bench_allreduce(m4a, comm, device, args):
+    n_elems = int(args.size_mib * 1024 * 1024 // 2)  # bf16 = 2 bytes
+    dtype = torch.bfloat16 if device.type == "cuda" else torch.bfloat16
+    t = torch.randn(n_elems, dtype=torch.float32, device=device).to(dtype)
+    t.requires_grad_()
+    grad_seed = torch.ones(n_elems, dtype=dtype, device=device)
+
+    def step():
+        t.grad = None
+        r = comm.Allreduce(t, m4a.MPI_SUM)
+        r.backward(grad_seed)
+
+    elapsed = time_steps(step, args.steps, args.warmup, device)
+    t_step = elapsed / args.steps
+    nbytes = n_elems * 2
+    return {
+        "metric": "allreduce_fwdbwd_algbw_GBps",
+        "value": comm.size * 2 * nbytes / t_step / 1e9,
+        "unit": "GB/s",
+        "config": {
+            "model": "allreduce_1GiB_bf16_fwd_bwd",
+            "tensor_mib": args.size_mib,
+            "global_batch": None,
+            "seq_len": None,
+            "parallelism": f"dp{comm.size}",
+        },
+    }, t_step
+
+
+def bench_linreg(m4a, comm, device, args):
+    # BASELINE config #3: data-parallel linear regression, per-rank local
+    # loss + allreduce (reference examples/simple_linear_regression.py)
+    from mpi4torch_amd.models.linreg import DistributedLinReg
+
+    n, f = 1 << 20, 32
+    model = DistributedLinReg(comm, n_features=f).to(device)
+    x = torch.randn(n, f, device=device)
+    y = torch.randn(n, device=device)
+    opt = torch.optim.SGD(model.parameters(), lr=1e-3)
+
+    def step():
+        opt.zero_grad()
+        loss = model.loss(x, y)
+        loss.backward()
+        opt.step()
+
+    elapsed = time_steps(step, args.steps, args.warmup, device)
+    t_step = elapsed / args.steps
+    return {
+        "metric": "linreg_steps_per_s",
+        "value": comm.size * args.steps / elapsed,
+        "unit": "steps/s*gpus",
+        "config": {
+            "model": "dp_linreg",
+            "global_batch": n * comm.size,
+            "seq_len": None,
+            "parallelism": f"dp{comm.size}",
+        },
+    }, t_step
+
+
+def bench_ring(m4a, comm, device, args):
+    # BASELINE config #4: 256 MiB ring exchange with autograd WaitHandles
+    n = 256 * 1024 * 1024 // 4  # fp32
+    t = torch.randn(n, device=device).requires_grad_()
+    grad_seed = torch.ones(n, device=device)
+    rank, world = comm.rank, comm.size
+
+    def step():
+        t.grad = None
+        req = comm.Isend(t, (rank + 1) % world, 0)
+        req2 = comm.Irecv(
+            m4a.JoinDummies(torch.empty_like(t), [req.dummy]),
+            (rank + world - 1) % world,
+            0,
+        )
+        res = comm.Wait(m4a.JoinDummiesHandle(req, [req2.dummy]))
+        res2 = comm.Wait(m4a.JoinDummiesHandle(req2, [res]))
+        res2.backward(grad_seed)
+
+    elapsed = time_steps(step, args.steps, args.warmup, device)
+    t_step = elapsed / args.steps
+    nbytes = n * 4
+    return {
+        "metric": "ring_exchange_fwdbwd_GBps",
+        "value": comm.size * 2 * nbytes / t_step / 1e9,
+        "unit": "GB/s",
+        "config": {
+            "model": "ring_256MiB_fp32_fwd_bwd",
+            "global_batch": None,
+            "seq_len": None,
+            "parallelism": f"ring{comm.size}",
+        },
+    }, t_step
+
+
+def bench_alltoall(m4a, comm, device, args):
+    # BASELINE config #5: 4096x4096 fp8 shards, expert-parallel pattern
+    world = comm.size
+    rows = 4096 * world
+    t32 = torch.randn(rows, 4096, device=device)
+    dtype = torch.float8_e4m3fn if device.type == "cuda" else torch.float8_e4m3fn
+    t = t32.to(dtype).requires_grad_()
+    grad_seed = torch.ones(4096, 4096 * world, device=device).to(dtype)
+
+    def step():
+        t.grad = None
+        r = comm.Alltoall(t, 1, 0, 4096)
+        r.backward(grad_seed)
+
+    elapsed = time_steps(step, args.steps, args.warmup, device)
+    t_step = elapsed / args.steps
+    nbytes = rows * 4096
+    return {
+        "metric": "alltoall_fp8_fwdbwd_GBps",
+        "value": comm.size * 2 * nbytes / t_step / 1e9,
+        "unit": "GB/s",
+        "config": {
+            "model": "alltoall_4096x4096_fp8",
+            "global_batch": None,
+            "seq_len": None,
+            "parallelism": f"ep{comm.size}",
+        },
+    }, t_step
+
+
+BENCHES = {
+    "allreduce": bench_allreduce,
+    "linreg": bench_linreg,
+    "ring": bench_ring,
+    "alltoall": bench_alltoall,
+}
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--config", default="allreduce", choices=sorted(BENCHES))
+    p.add_argument("--size-mib", type=float, default=1024.0,
+                   help="allreduce tensor size per GPU (MiB)")
+    args = p.parse_args()
+
+    m4a, rank, world, device = setup_world(args)
+    comm = m4a.COMM_WORLD
+    assert comm.size == world
+
+    result, t_step = BENCHES[args.config](m4a, comm, device, args)
+    if rank == 0:
+        out = {
+            "metric": result["metric"],
+            "value": round(result["value"], 3),
+            "unit": result["unit"],
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(t_step * 1e3, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # the reference publishes no numbers
+            "dtype": "bf16" if args.config == "allreduce" else
+                     ("fp8" if args.config == "alltoall" else "fp32"),
+            "data": "synthetic",
+            "config": result["config"],
+        }
+        print(json.dumps(out), flush=True)
+
+    import torch.distributed as dist
+
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

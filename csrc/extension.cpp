@@ -50,6 +50,7 @@ static auto communicator_class =
         .def("Allgather", &Communicator::Allgather)
         .def("Scatter", &Communicator::Scatter)
         .def("Alltoall", &Communicator::Alltoall)
+        .def("Iallreduce", &Communicator::Iallreduce)
         .def("Isend", &Communicator::Isend)
         .def("Irecv", &Communicator::Irecv)
         .def("Wait", &Communicator::Wait)
@@ -88,6 +89,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         [](bool enabled) { m4a::config().force_host_staging = enabled; });
   m.def("host_staging_forced",
         []() { return m4a::config().force_host_staging; });
+
+  m.def("_pack_roundtrip", &m4a::debug_pack_roundtrip);
+  m.def("_bitwise_reduce", &m4a::debug_bitwise_reduce);
 
   m.def("_rccl_version", []() {
     int v = 0;

@@ -997,6 +997,7 @@ namespace {
 
 constexpr int64_t kIsendOp = 0;
 constexpr int64_t kIrecvOp = 1;
+constexpr int64_t kIallreduceOp = 2;
 
 double ptr_hash(const void* p) {
   return (double)(0xFFFFFFFFull & std::hash<const void*>()(p));
@@ -1046,6 +1047,8 @@ variable_list WaitBackward::apply(variable_list&& grads) {
               "but is ",
               next_node ? next_node->name() : "<null>",
               ". Only JoinDummiesHandle may be applied to a wait handle.");
+  TORCH_CHECK(op != kIallreduceOp,
+              "mpi4torch_amd: Iallreduce handles carry no gradient");
   if (op == kIsendOp) {
     // adjoint of send is receive (on the dedicated backward channel)
     auto buf = at::zeros(buf_sizes, buf_options);
@@ -1126,6 +1129,35 @@ std::vector<at::Tensor> Communicator::IrecvImpl(const Tensor& input,
   return result;
 }
 
+std::vector<at::Tensor> Communicator::Iallreduce(const Tensor& input,
+                                                 int64_t op) {
+  check_op(op);
+  TORCH_CHECK(!torch::autograd::compute_requires_grad(input),
+              "mpi4torch_amd: Iallreduce does not support autograd; use "
+              "Allreduce, or detach the input (gradient buckets are already "
+              "grad tensors)");
+  at::AutoDispatchBelowADInplaceOrView guard;
+  DeviceStager stager(input);
+  auto in = stager.to_comm(input).contiguous().variable_data();
+  auto& tr = tr_for(in);
+  TORCH_CHECK(is_arith(op) && native_reduce_dtype(tr, in.scalar_type()),
+              "mpi4torch_amd: Iallreduce supports native arithmetic "
+              "reductions only (op ", red_op_name(op), ", dtype ",
+              in.scalar_type(), ")");
+  auto out = at::empty_like(in);
+  uint64_t req = tr.iallreduce(in, out, (RedOp)op);
+  auto meta = at::empty({7}, at::TensorOptions().dtype(at::kDouble));
+  auto* m = meta.data_ptr<double>();
+  m[0] = (double)req;
+  m[1] = (double)kIallreduceOp;
+  m[2] = 0.0;
+  m[3] = 0.0;
+  m[4] = ptr_hash(out.data_ptr());
+  m[5] = (double)(int64_t)stager.orig_device_.type();
+  m[6] = (double)stager.orig_device_.index();
+  return {meta, out, in};
+}
+
 std::vector<at::Tensor> Communicator::Isend(const Tensor& input, int64_t dest,
                                             int64_t tag) {
   return IsendImpl(input, dest, tag, /*backward_channel=*/false);
@@ -1171,6 +1203,7 @@ Tensor Communicator::Wait(const std::vector<Tensor>& handle) {
     if (op == kIsendOp) {
       return handle[2].variable_data();
     }
+    // kIrecvOp and kIallreduceOp both return the (possibly unstaged) buffer
     auto buf = handle[1];
     if (buf.device() != orig_device) {
       buf = buf.to(orig_device);
@@ -1179,6 +1212,39 @@ Tensor Communicator::Wait(const std::vector<Tensor>& handle) {
   }();
   attach_history(result, grad_fn);
   return result;
+}
+
+
+// ---------------------------------------------------------------------------
+// Debug/test entry points: exercise the CDNA4 kernels directly on one GPU
+// (world-size-independent numerics tests; see tests/test_gpu.py).
+// ---------------------------------------------------------------------------
+
+Tensor debug_pack_roundtrip(const Tensor& input, int64_t axis,
+                            std::vector<int64_t> counts) {
+  auto in = input.contiguous();
+  axis = at::maybe_wrap_dim(axis, in.dim());
+  const auto g = axis_geom(in, axis);
+  auto displs = prefix_displs(counts);
+  TORCH_CHECK(displs.back() + counts.back() == g.axis,
+              "counts must partition the axis");
+  auto blocks = make_blocks(in, g.before, g.after, counts, displs, nullptr);
+  move_axis_blocks(in, axis, displs, counts, blocks, /*pack=*/true);
+  auto out = at::zeros_like(in);
+  move_axis_blocks(out, axis, displs, counts, blocks, /*pack=*/false);
+  return out;
+}
+
+Tensor debug_bitwise_reduce(const Tensor& stacked, int64_t op) {
+  TORCH_CHECK(stacked.is_cuda(), "debug_bitwise_reduce is a GPU kernel test");
+  TORCH_CHECK(stacked.dim() >= 2);
+  auto in = stacked.contiguous();
+  const int64_t n = in.size(0);
+  auto out = at::empty(in.sizes().slice(1).vec(), in.options());
+  const int64_t chunk_bytes = out.numel() * out.element_size();
+  launch_bitwise_reduce(in.data_ptr(), out.data_ptr(), chunk_bytes, (int)n,
+                        (int)op, current_gpu_stream(in));
+  return out;
 }
 
 } // namespace m4a

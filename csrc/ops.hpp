@@ -37,6 +37,12 @@ struct Communicator : torch::CustomClassHolder {
   at::Tensor Alltoall(const at::Tensor& input, int64_t gatheraxis,
                       int64_t scatteraxis, int64_t numelem);
 
+  // Non-blocking allreduce (MI355X-first overlap primitive, not in the
+  // reference API): returns a wait handle; no autograd through it — use
+  // Allreduce for differentiable paths. Gradient bucketing (parallel/ddp)
+  // is its main consumer.
+  std::vector<at::Tensor> Iallreduce(const at::Tensor& input, int64_t op);
+
   // Non-blocking p2p. Handle contract identical to the reference
   // (csrc/extension.cpp:1094-1107): [meta tensor, comm buffer, input].
   std::vector<at::Tensor> Isend(const at::Tensor& input, int64_t dest,
@@ -71,5 +77,11 @@ struct Communicator : torch::CustomClassHolder {
 // JoinDummies free op (reference csrc/extension.cpp:989-1046).
 at::Tensor join_dummies(const at::Tensor& loopthrough,
                         const std::vector<at::Tensor>& dummies);
+
+// Debug/test entry points (tests/test_gpu.py): exercise the CDNA4 slab and
+// bitwise-reduce kernels directly, without a multi-rank world.
+at::Tensor debug_pack_roundtrip(const at::Tensor& input, int64_t axis,
+                                std::vector<int64_t> counts);
+at::Tensor debug_bitwise_reduce(const at::Tensor& stacked, int64_t op);
 
 } // namespace m4a

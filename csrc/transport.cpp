@@ -314,6 +314,20 @@ struct LocalTransport final : Transport {
     TORCH_CHECK(peer == 0, "world_size is 1; cannot receive from rank ", peer);
     return self_matcher().irecv(buf, tag, ch);
   }
+  uint64_t iallreduce(const at::Tensor& in, at::Tensor& out,
+                      RedOp) override {
+    out.copy_(in, true);
+    Request r;
+    r.buffer = out;
+    if (out.is_cuda()) {
+      r.gpu = true;
+      r.device = (int)out.get_device();
+      auto cur = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(r.device);
+      r.event = EventPool::forDevice(r.device).acquire();
+      M4A_HIP_CHECK(hipEventRecord(r.event, cur.stream()));
+    }
+    return requests().add(std::move(r));
+  }
 };
 
 // ---------------------------------------------------------------------------
@@ -429,6 +443,18 @@ struct C10dTransport final : Transport {
     Request r;
     r.buffer = buf;
     r.work = backend_->recv(ts, peer, user_tag(tag, ch));
+    return requests().add(std::move(r));
+  }
+  uint64_t iallreduce(const at::Tensor& in, at::Tensor& out,
+                      RedOp op) override {
+    std::lock_guard<std::mutex> g(mu_);
+    out.copy_(in);
+    std::vector<at::Tensor> ts{out};
+    c10d::AllreduceOptions opts;
+    opts.reduceOp = c10d_red_op(op);
+    Request r;
+    r.buffer = out;
+    r.work = backend_->allreduce(ts, opts);
     return requests().add(std::move(r));
   }
 
@@ -596,6 +622,19 @@ struct RcclTransport final : Transport {
     M4A_NCCL_CHECK(ncclRecv(buf.data_ptr(), nbytes(buf), ncclUint8, peer,
                             comm(ch), stream(ch)));
     return make_gpu_request(ch, buf);
+  }
+  uint64_t iallreduce(const at::Tensor& in, at::Tensor& out,
+                      RedOp op) override {
+    // On the collective stream WITHOUT the tail hop: the caller's stream
+    // only waits when the returned request is waited — this is what lets
+    // gradient-bucket allreduce overlap the rest of backward.
+    std::lock_guard<std::mutex> g(mu_);
+    enter_side(Channel::Coll, {in, out});
+    M4A_NCCL_CHECK(ncclAllReduce(in.data_ptr(), out.data_ptr(), in.numel(),
+                                 nccl_reduce_dtype(in.scalar_type()),
+                                 nccl_red_op(op), comm(Channel::Coll),
+                                 stream(Channel::Coll)));
+    return make_gpu_request(Channel::Coll, out);
   }
 
  private:

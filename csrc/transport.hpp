@@ -67,6 +67,10 @@ struct Transport {
   virtual uint64_t isend(const at::Tensor& buf, int peer, int tag,
                          Channel ch) = 0;
   virtual uint64_t irecv(at::Tensor& buf, int peer, int tag, Channel ch) = 0;
+  // Non-blocking allreduce (no autograd; overlap primitive for gradient
+  // bucketing — an MI355X-first extension beyond the reference's API).
+  virtual uint64_t iallreduce(const at::Tensor& in, at::Tensor& out,
+                              RedOp op) = 0;
 };
 
 // Completes a request: GPU → inserts a wait into the caller's current

@@ -182,6 +182,13 @@ class MPI_Communicator:
         """
         return self._comm.Alltoall(tensor, gatheraxis, scatteraxis, numelem)
 
+    def Iallreduce(self, tensor: torch.Tensor, op: int) -> WaitHandle:
+        """Non-blocking Allreduce (no autograd): returns a WaitHandle whose
+        Wait() yields the reduced tensor. The overlap primitive behind
+        gradient bucketing (mpi4torch_amd.parallel.DistributedDataParallel);
+        not part of the reference API."""
+        return WaitHandle(self._comm.Iallreduce(tensor, op))
+
     def Isend(self, tensor: torch.Tensor, dest: int, tag: int) -> WaitHandle:
         """Non-blocking send; complete with Wait. Backward: reverse recv."""
         return WaitHandle(self._comm.Isend(tensor, dest, tag))

@@ -1,0 +1,10 @@
+"""Example/benchmark models built on the mpi4torch_amd primitives.
+
+The reference ships no model zoo — its "models" are the documented
+data-parallel patterns (reference examples/simple_linear_regression.py,
+doc/examples.rst). These modules are those patterns as reusable code.
+"""
+
+from .linreg import DistributedLinReg
+
+__all__ = ["DistributedLinReg"]

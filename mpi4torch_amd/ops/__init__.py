@@ -1,0 +1,70 @@
+"""Functional wrappers over the COMM_WORLD communicator.
+
+Thin conveniences for code that does not want to thread a communicator
+object through call sites; every function accepts an explicit ``comm=``.
+"""
+
+from typing import List, Optional
+
+import torch
+
+import mpi4torch_amd as _m
+
+
+def _comm(comm):
+    return comm if comm is not None else _m.COMM_WORLD
+
+
+def allreduce(tensor: torch.Tensor, op: Optional[int] = None, comm=None):
+    return _comm(comm).Allreduce(tensor, _m.MPI_SUM if op is None else op)
+
+
+def bcast_(tensor: torch.Tensor, root: int = 0, comm=None):
+    return _comm(comm).Bcast_(tensor, root)
+
+
+def reduce_(tensor: torch.Tensor, op: Optional[int] = None, root: int = 0,
+            comm=None):
+    return _comm(comm).Reduce_(tensor, _m.MPI_SUM if op is None else op, root)
+
+
+def gather(tensor: torch.Tensor, axis: int = 0, root: int = 0, comm=None):
+    return _comm(comm).Gather(tensor, axis, root)
+
+
+def allgather(tensor: torch.Tensor, axis: int = 0, comm=None):
+    return _comm(comm).Allgather(tensor, axis)
+
+
+def scatter(tensor: torch.Tensor, axis: int = 0, numelem: int = 1,
+            root: int = 0, comm=None):
+    return _comm(comm).Scatter(tensor, axis, numelem, root)
+
+
+def alltoall(tensor: torch.Tensor, gatheraxis: int, scatteraxis: int,
+             numelem: int, comm=None):
+    return _comm(comm).Alltoall(tensor, gatheraxis, scatteraxis, numelem)
+
+
+def isend(tensor: torch.Tensor, dest: int, tag: int = 0, comm=None):
+    return _comm(comm).Isend(tensor, dest, tag)
+
+
+def irecv(tensor: torch.Tensor, source: int, tag: int = 0, comm=None):
+    return _comm(comm).Irecv(tensor, source, tag)
+
+
+def wait(handle, comm=None):
+    return _comm(comm).Wait(handle)
+
+
+def send(tensor: torch.Tensor, dest: int, tag: int = 0, comm=None):
+    return _comm(comm).Send(tensor, dest, tag)
+
+
+def recv(tensor: torch.Tensor, source: int, tag: int = 0, comm=None):
+    return _comm(comm).Recv(tensor, source, tag)
+
+
+def join_dummies(loopthrough: torch.Tensor, dummies: List[torch.Tensor]):
+    return _m.JoinDummies(loopthrough, dummies)

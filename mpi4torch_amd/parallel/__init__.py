@@ -1,0 +1,17 @@
+"""Parallelism strategies as thin layers over the primitives.
+
+The reference is a primitive library; its docs demonstrate data parallelism
+(doc/examples.rst) and its axis-aware Alltoall IS the Ulysses sequence<->
+head reshard primitive. These modules make those patterns
+reusable, tuned for one-process-per-GPU RCCL over xGMI.
+"""
+
+from .ddp import DistributedDataParallel
+from .ulysses import ulysses_reshard, seq_to_head, head_to_seq
+
+__all__ = [
+    "DistributedDataParallel",
+    "ulysses_reshard",
+    "seq_to_head",
+    "head_to_seq",
+]

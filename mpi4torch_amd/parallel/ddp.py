@@ -1,0 +1,134 @@
+"""Bucketed data-parallel gradient synchronization over xGMI.
+
+A light DDP built on mpi4torch_amd's non-blocking Allreduce: gradients are
+flattened into buckets as backward produces them and each full bucket's
+allreduce is launched immediately on the collective stream, overlapping the
+rest of backward. Waits are deferred to finish_gradient_sync() (call it
+between loss.backward() and optimizer.step()).
+
+Sizing rationale (MI355X): intra-node xGMI is 7 point-to-point links per
+GPU at ~153 GB/s each; ring allreduce is per-link bound, and RCCL needs
+buckets large enough to stripe across its channels — default 128 MiB
+(larger than CUDA-era defaults; 288 GB HBM3E makes big buckets free).
+"""
+
+from typing import List, Optional
+
+import torch
+
+import mpi4torch_amd as m4a
+
+
+class _Bucket:
+    def __init__(self, params: List[torch.nn.Parameter]):
+        self.params = params
+        self.pending = 0
+        self.flat: Optional[torch.Tensor] = None
+        self.handle = None
+
+
+class DistributedDataParallel(torch.nn.Module):
+    """Wrap a module for data-parallel training.
+
+    Usage::
+
+        model = DistributedDataParallel(module)
+        loss = model(x).sum()
+        loss.backward()
+        model.finish_gradient_sync()   # waits overlapped allreduces
+        optimizer.step()
+    """
+
+    def __init__(self, module: torch.nn.Module, comm=None,
+                 bucket_cap_mb: int = 128, average: bool = True):
+        super().__init__()
+        self.module = module
+        self.comm = comm if comm is not None else m4a.COMM_WORLD
+        self.average = average
+        self._sync_enabled = True
+
+        # broadcast initial parameters from rank 0 so replicas agree
+        if self.comm.size > 1:
+            with torch.no_grad():
+                for p in self.module.parameters():
+                    self.comm.Bcast_(p.data, 0)
+
+        # buckets in reverse parameter order (grads arrive roughly in
+        # reverse forward order during backward)
+        params = [p for p in self.module.parameters() if p.requires_grad]
+        cap = bucket_cap_mb * 1024 * 1024
+        self._buckets: List[_Bucket] = []
+        cur: List[torch.nn.Parameter] = []
+        size = 0
+        for p in reversed(params):
+            cur.append(p)
+            size += p.numel() * p.element_size()
+            if size >= cap:
+                self._buckets.append(_Bucket(cur))
+                cur, size = [], 0
+        if cur:
+            self._buckets.append(_Bucket(cur))
+
+        self._param_bucket = {}
+        for b in self._buckets:
+            for p in b.params:
+                self._param_bucket[p] = b
+        for p in params:
+            p.register_post_accumulate_grad_hook(self._grad_ready)
+
+        self._reset_pending()
+
+    def _reset_pending(self):
+        for b in self._buckets:
+            b.pending = len(b.params)
+            b.flat = None
+            b.handle = None
+
+    def _grad_ready(self, p: torch.nn.Parameter):
+        if not self._sync_enabled or self.comm.size == 1:
+            return
+        b = self._param_bucket[p]
+        b.pending -= 1
+        if b.pending == 0:
+            with torch.no_grad():
+                b.flat = torch.cat(
+                    [q.grad.reshape(-1) for q in b.params]
+                ).contiguous()
+                # non-blocking: overlaps the rest of backward
+                b.handle = self.comm.Iallreduce(b.flat, m4a.MPI_SUM)
+
+    def forward(self, *args, **kwargs):
+        return self.module(*args, **kwargs)
+
+    def finish_gradient_sync(self):
+        """Wait all in-flight bucket allreduces and scatter results back
+        into .grad. Call between backward() and optimizer.step()."""
+        if self.comm.size == 1:
+            return
+        scale = 1.0 / self.comm.size if self.average else 1.0
+        with torch.no_grad():
+            for b in self._buckets:
+                if b.handle is None:
+                    continue
+                reduced = self.comm.Wait(b.handle)
+                if self.average:
+                    reduced = reduced * scale
+                off = 0
+                for q in b.params:
+                    n = q.numel()
+                    q.grad.copy_(reduced[off : off + n].view_as(q.grad))
+                    off += n
+        self._reset_pending()
+
+    def no_sync(self):
+        """Context manager: skip gradient sync (gradient accumulation)."""
+        ddp = self
+
+        class _NoSync:
+            def __enter__(self):
+                ddp._sync_enabled = False
+
+            def __exit__(self, *exc):
+                ddp._sync_enabled = True
+
+        return _NoSync()

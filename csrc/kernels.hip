@@ -11,6 +11,15 @@
 // workgroups to fill all 8 XCDs. Pure streaming copies have no inter-block
 // reuse, so no XCD-aware blockIdx remap is needed (it only pays when
 // neighboring blocks share operand panels — cdna_hip_programming.md T1).
+//
+// Slab geometry: a slab is [before] rows of [row_b] contiguous bytes each
+// (the (count, after) dims of an axis slice collapse into one row — both
+// sides advance by after_b per axis element, so consecutive axis elements
+// are contiguous on BOTH sides). Rows are vectorized at the widest granule
+// whose alignment PHASE matches between src and dst ((src - dst) % G == 0):
+// each row gets a bytewise head up to dst 16/4/2-alignment, a granule body,
+// and a bytewise tail, so odd element counts and odd displacements still
+// stream at near-HBM rate instead of falling to a byte loop.
 
 #include <hip/hip_runtime.h>
 #include "kernels.hpp"
@@ -19,16 +28,14 @@ namespace m4a {
 
 namespace {
 
-// Slab geometry precomputed in *chunks* (one chunk = Granule bytes).
 struct SlabArgs {
   const char* src;
   char* dst;
-  long long chunks_per_row;   // after_b / granule
-  long long rows_per_b;       // count
-  long long total_chunks;     // before * count * chunks_per_row
+  long long row_b;          // bytes per row (collapsed count*after)
+  long long chunks_per_row; // body granules + 1 (head/tail chunk)
+  long long total_chunks;   // before * chunks_per_row
   long long src_pitch_b;
   long long dst_pitch_b;
-  long long after_b;
 };
 
 template <int N>
@@ -36,34 +43,40 @@ struct SlabPack {
   SlabArgs s[N];
 };
 
+// One chunk index c in [0, chunks_per_row): c < n_body copies granule c of
+// the phase-aligned body; c == n_body copies the head and tail bytes.
 template <typename VecT, int NSLABS>
 __global__ __launch_bounds__(256) void slab_copy_kernel(SlabPack<NSLABS> pack) {
+  constexpr long long G = (long long)sizeof(VecT);
   const SlabArgs& a = pack.s[blockIdx.z];
   const long long stride = (long long)gridDim.x * blockDim.x;
   for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
        i < a.total_chunks; i += stride) {
-    // i -> (b, c, k): row-major over [before][count][chunks_per_row]
-    const long long row = i / a.chunks_per_row;       // = b*count + c
-    const long long k = i - row * a.chunks_per_row;
-    const long long b = row / a.rows_per_b;
-    const long long c = row - b * a.rows_per_b;
-    const long long off = c * a.after_b + k * (long long)sizeof(VecT);
-    const VecT* sp =
-        reinterpret_cast<const VecT*>(a.src + b * a.src_pitch_b + off);
-    VecT* dp = reinterpret_cast<VecT*>(a.dst + b * a.dst_pitch_b + off);
-    *dp = *sp;
+    const long long b = i / a.chunks_per_row;
+    const long long c = i - b * a.chunks_per_row;
+    const char* srow = a.src + b * a.src_pitch_b;
+    char* drow = a.dst + b * a.dst_pitch_b;
+    // head: bytes until drow reaches G alignment (phase equality makes srow
+    // aligned at the same point); clamped for rows shorter than the head
+    long long head =
+        (G - ((long long)(uintptr_t)drow & (G - 1))) & (G - 1);
+    if (head > a.row_b) head = a.row_b;
+    const long long body = (a.row_b - head) / G;
+    if (c < body) {
+      const VecT* sp = reinterpret_cast<const VecT*>(srow + head) + c;
+      VecT* dp = reinterpret_cast<VecT*>(drow + head) + c;
+      *dp = *sp;
+    } else {
+      // head + tail bytes, done by the one extra chunk per row
+      for (long long k = 0; k < head; ++k) drow[k] = srow[k];
+      for (long long k = head + body * G; k < a.row_b; ++k) drow[k] = srow[k];
+    }
   }
-}
-
-inline bool aligned_to(const SlabDesc& d, int64_t g) {
-  return (d.after_b % g == 0) &&
-         (reinterpret_cast<uintptr_t>(d.src) % g == 0) &&
-         (reinterpret_cast<uintptr_t>(d.dst) % g == 0) &&
-         (d.src_pitch_b % g == 0) && (d.dst_pitch_b % g == 0);
 }
 
 template <typename VecT>
 void launch_bucket(const SlabDesc* descs, int n, hipStream_t stream) {
+  constexpr long long G = (long long)sizeof(VecT);
   while (n > 0) {
     const int take = n < kMaxSlabsPerLaunch ? n : kMaxSlabsPerLaunch;
     SlabPack<kMaxSlabsPerLaunch> pack{};
@@ -73,12 +86,13 @@ void launch_bucket(const SlabDesc* descs, int n, hipStream_t stream) {
       SlabArgs& a = pack.s[i];
       a.src = static_cast<const char*>(d.src);
       a.dst = static_cast<char*>(d.dst);
-      a.chunks_per_row = d.after_b / (int64_t)sizeof(VecT);
-      a.rows_per_b = d.count;
-      a.total_chunks = d.before * d.count * a.chunks_per_row;
+      a.row_b = d.count * d.after_b;  // collapsed row
       a.src_pitch_b = d.src_pitch_b;
       a.dst_pitch_b = d.dst_pitch_b;
-      a.after_b = d.after_b;
+      // worst-case head is G-1 bytes; one extra chunk per row covers
+      // head+tail. Rows shorter than G go entirely through that chunk.
+      a.chunks_per_row = (a.row_b >= G ? a.row_b / G : 0) + 1;
+      a.total_chunks = d.before * a.chunks_per_row;
       if (a.total_chunks > max_chunks) max_chunks = a.total_chunks;
     }
     for (int i = take; i < kMaxSlabsPerLaunch; ++i) {
@@ -98,6 +112,18 @@ void launch_bucket(const SlabDesc* descs, int n, hipStream_t stream) {
     descs += take;
     n -= take;
   }
+}
+
+// Widest granule whose alignment phase matches on both sides for every row:
+// needs (src - dst) % G == 0 and (pitch difference) % G == 0.
+inline int slab_granule(const SlabDesc& d) {
+  const long long delta =
+      (long long)((uintptr_t)d.src - (uintptr_t)d.dst);
+  const long long pdelta = d.src_pitch_b - d.dst_pitch_b;
+  for (int g = 16; g > 1; g >>= 1) {
+    if ((delta & (g - 1)) == 0 && (pdelta & (g - 1)) == 0) return g;
+  }
+  return 1;
 }
 
 struct OpAnd {
@@ -155,27 +181,19 @@ void launch_bitred(const void* in, void* out, int64_t chunk_elems, int nranks,
 } // namespace
 
 void launch_slab_copy(const SlabDesc* descs, int n, hipStream_t stream) {
-  // Bucket consecutive slabs by the widest granule they admit so each launch
-  // is uniform. In practice all slabs of one collective share alignment.
+  // Bucket consecutive slabs by alignment-phase granule so each launch is
+  // uniform. In practice all slabs of one collective share a granule.
   int i = 0;
   while (i < n) {
-    const int64_t g = aligned_to(descs[i], 16) ? 16
-                      : aligned_to(descs[i], 4) ? 4
-                                                : 1;
+    const int g = slab_granule(descs[i]);
     int j = i + 1;
-    while (j < n) {
-      const int64_t gj = aligned_to(descs[j], 16) ? 16
-                         : aligned_to(descs[j], 4) ? 4
-                                                   : 1;
-      if (gj != g) break;
-      ++j;
-    }
-    if (g == 16) {
-      launch_bucket<uint4>(descs + i, j - i, stream);
-    } else if (g == 4) {
-      launch_bucket<unsigned int>(descs + i, j - i, stream);
-    } else {
-      launch_bucket<unsigned char>(descs + i, j - i, stream);
+    while (j < n && slab_granule(descs[j]) == g) ++j;
+    switch (g) {
+      case 16: launch_bucket<uint4>(descs + i, j - i, stream); break;
+      case 8: launch_bucket<unsigned long long>(descs + i, j - i, stream); break;
+      case 4: launch_bucket<unsigned int>(descs + i, j - i, stream); break;
+      case 2: launch_bucket<unsigned short>(descs + i, j - i, stream); break;
+      default: launch_bucket<unsigned char>(descs + i, j - i, stream); break;
     }
     i = j;
   }

@@ -15,7 +15,11 @@ namespace m4a {
 // Copies, for b in [0,before), c in [0,count):
 //   dst[b*dst_pitch_b + c*after_b + 0..after_b) =
 //   src[b*src_pitch_b + c*after_b + 0..after_b)
-// All quantities in BYTES except `before`/`count` (row counts).
+// All quantities in BYTES except `before`/`count` (row counts). Note the
+// c-stride is after_b on BOTH sides by definition, so (count, after)
+// collapse into one contiguous row of count*after_b bytes per b — the
+// kernel exploits this and vectorizes rows with phase-aligned 16/8/4/2-byte
+// granules plus bytewise head/tail.
 struct SlabDesc {
   const void* src;
   void* dst;

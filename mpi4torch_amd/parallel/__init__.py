@@ -8,10 +8,16 @@ reusable, tuned for one-process-per-GPU RCCL over xGMI.
 
 from .ddp import DistributedDataParallel
 from .ulysses import ulysses_reshard, seq_to_head, head_to_seq
+from .tp import (ColumnParallelLinear, RowParallelLinear, TensorParallelMLP,
+                 copy_to_tp_region)
 
 __all__ = [
     "DistributedDataParallel",
     "ulysses_reshard",
     "seq_to_head",
     "head_to_seq",
+    "ColumnParallelLinear",
+    "RowParallelLinear",
+    "TensorParallelMLP",
+    "copy_to_tp_region",
 ]

@@ -644,10 +644,29 @@ struct RcclTransport final : Transport {
   ncclComm_t comm(Channel ch) { return comms_[(int)ch]; }
   hipStream_t stream(Channel ch) { return streams_[(int)ch].stream(); }
 
+  // Failure detection (the reference has none beyond ierr
+  // checks): surface asynchronous RCCL errors (peer crash, xGMI fault,
+  // aborted communicator) as exceptions at the next collective instead of
+  // hanging the job.
+  void check_async_errors() {
+    for (int i = 0; i < 3; ++i) {
+      if (!comms_[i]) continue;
+      ncclResult_t async_err = ncclSuccess;
+      if (ncclCommGetAsyncError(comms_[i], &async_err) == ncclSuccess) {
+        TORCH_CHECK(async_err == ncclSuccess || async_err == ncclInProgress,
+                    "mpi4torch_amd: RCCL communicator (channel ", i,
+                    ") is in an error state: ",
+                    ncclGetErrorString(async_err),
+                    " — a peer likely failed; restart the job");
+      }
+    }
+  }
+
   // Make the side stream wait on the caller's current stream, and record
   // every touched tensor with the caching allocator against the side
   // stream so its memory is not reused while the collective is in flight.
   void enter_side(Channel ch, const std::vector<at::Tensor>& tensors) {
+    check_async_errors();
     auto cur = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(device_);
     auto& pool = EventPool::forDevice(device_);
     hipEvent_t e = pool.acquire();

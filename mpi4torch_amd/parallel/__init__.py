@@ -10,6 +10,7 @@ from .ddp import DistributedDataParallel
 from .ulysses import ulysses_reshard, seq_to_head, head_to_seq
 from .tp import (ColumnParallelLinear, RowParallelLinear, TensorParallelMLP,
                  copy_to_tp_region)
+from .pipeline import GPipe
 
 __all__ = [
     "DistributedDataParallel",
@@ -20,4 +21,5 @@ __all__ = [
     "RowParallelLinear",
     "TensorParallelMLP",
     "copy_to_tp_region",
+    "GPipe",
 ]

@@ -217,8 +217,12 @@ bool native_reduce_dtype(const Transport& tr, at::ScalarType t) {
       case at::kDouble:
       case at::kBFloat16:
         return true;
+      case at::kFloat8_e4m3fn:
+      case at::kFloat8_e5m2:
+        // native when this RCCL reduces fp8 (runtime probe), else upcast
+        return const_cast<Transport&>(tr).fp8_reduce_supported(t);
       default:
-        return false;  // fp8/short/bool go through the upcast path
+        return false;  // short/bool go through the upcast path
     }
   }
   // gloo (like RCCL) has no int16: kShort takes the int32 upcast path

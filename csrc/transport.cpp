@@ -623,6 +623,27 @@ struct RcclTransport final : Transport {
                             comm(ch), stream(ch)));
     return make_gpu_request(ch, buf);
   }
+  // Runtime probe: does this RCCL build reduce fp8 natively? One tiny
+  // allreduce per dtype, issued symmetrically on all ranks (the op layer
+  // reaches this on every rank at the same program point). Cached.
+  bool fp8_reduce_supported(at::ScalarType t) override {
+    const int idx = (t == at::kFloat8_e4m3fn) ? 0 : 1;
+    std::lock_guard<std::mutex> g(mu_);
+    if (fp8_state_[idx] != 0) return fp8_state_[idx] > 0;
+    void* buf = nullptr;
+    M4A_HIP_CHECK(hipMalloc(&buf, 64));
+    M4A_HIP_CHECK(hipMemsetAsync(buf, 0, 64, stream(Channel::Coll)));
+    ncclResult_t rc = ncclAllReduce(
+        buf, static_cast<char*>(buf) + 32, 32, nccl_reduce_dtype(t), ncclSum,
+        comm(Channel::Coll), stream(Channel::Coll));
+    if (rc == ncclSuccess) {
+      M4A_HIP_CHECK(hipStreamSynchronize(stream(Channel::Coll)));
+    }
+    M4A_HIP_CHECK(hipFree(buf));
+    fp8_state_[idx] = (rc == ncclSuccess) ? 1 : -1;
+    return fp8_state_[idx] > 0;
+  }
+
   uint64_t iallreduce(const at::Tensor& in, at::Tensor& out,
                       RedOp op) override {
     // On the collective stream WITHOUT the tail hop: the caller's stream
@@ -712,6 +733,7 @@ struct RcclTransport final : Transport {
   int rank_ = 0, size_ = 1;
   std::mutex mu_;
   ncclComm_t comms_[3] = {nullptr, nullptr, nullptr};
+  int fp8_state_[2] = {0, 0};  // 0 unknown, 1 native, -1 cast fallback
   c10::hip::HIPStreamMasqueradingAsCUDA streams_[3];
 };
 

@@ -1,5 +1,6 @@
 from .timing import CollectiveTimer, algbw_gbps, busbw_gbps
 from .tracing import trace, TracedCommunicator, TraceRecord
+from .checkpoint import save_checkpoint, load_checkpoint
 
 __all__ = [
     "CollectiveTimer",
@@ -8,4 +9,6 @@ __all__ = [
     "trace",
     "TracedCommunicator",
     "TraceRecord",
+    "save_checkpoint",
+    "load_checkpoint",
 ]

@@ -32,3 +32,37 @@ def test_busbw_math():
     assert algbw_gbps(1e9, 1.0) == 1.0
     assert abs(busbw_gbps(1e9, 1.0, 8, "allreduce") - 2 * 7 / 8) < 1e-9
     assert busbw_gbps(1e9, 1.0, 1) == 1.0
+
+
+def _ckpt_worker(rank, world, tmpdir):
+    import os
+    import torch
+    import mpi4torch_amd as m
+    from mpi4torch_amd.utils import save_checkpoint, load_checkpoint
+
+    comm = m.COMM_WORLD
+    torch.manual_seed(5)
+    net = torch.nn.Linear(4, 3)
+    opt = torch.optim.SGD(net.parameters(), lr=0.1)
+    net(torch.randn(2, 4)).sum().backward()
+    opt.step()
+    path = os.path.join(tmpdir, "ckpt.pt")
+    save_checkpoint(path, net, opt, extra={"step": 7})
+    # ranks see the file via the shared filesystem; barrier via collective
+    comm.Allreduce(torch.zeros(1), m.MPI_SUM)
+
+    torch.manual_seed(1000 + rank)  # diverged fresh model per rank
+    net2 = torch.nn.Linear(4, 3)
+    opt2 = torch.optim.SGD(net2.parameters(), lr=0.1)
+    extra = load_checkpoint(path, net2, opt2)
+    if rank == 0:
+        assert extra == {"step": 7}
+    for p, q in zip(net.parameters(), net2.parameters()):
+        assert torch.allclose(p, q), "checkpoint round-trip diverged"
+
+
+def test_checkpoint_ws2(tmp_path_factory):
+    import tempfile
+
+    d = tempfile.mkdtemp()
+    run_spmd(2, _ckpt_worker, d)

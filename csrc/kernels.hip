@@ -22,6 +22,7 @@
 // stream at near-HBM rate instead of falling to a byte loop.
 
 #include <hip/hip_runtime.h>
+#include <hip/hip_fp8.h>
 #include "kernels.hpp"
 
 namespace m4a {
@@ -223,6 +224,137 @@ void launch_bitwise_reduce(const void* in, void* out, int64_t chunk_bytes,
         launch_bitred<unsigned char, OpXor>(in, out, chunk_bytes, nranks, stream);
         break;
     }
+  }
+}
+
+
+namespace {
+
+// OCP fp8 <-> f32 through the gfx950 hardware conversion ops exposed by
+// hip_fp8.h (__hip_fp8_e4m3 / __hip_fp8_e5m2 are the OCP encodings —
+// NOT the MI300X fnuz variants; cdna_hip_programming.md §4).
+template <bool E5M2>
+__device__ inline float fp8_to_f32(unsigned char b) {
+  if (E5M2) {
+    __hip_fp8_e5m2 v;
+    v.__x = b;
+    return float(v);
+  }
+  __hip_fp8_e4m3 v;
+  v.__x = b;
+  return float(v);
+}
+
+template <bool E5M2>
+__device__ inline unsigned char f32_to_fp8(float f) {
+  if (E5M2) {
+    __hip_fp8_e5m2 v(f);
+    return v.__x;
+  }
+  __hip_fp8_e4m3 v(f);
+  return v.__x;
+}
+
+struct FAdd {
+  __device__ static float apply(float a, float b) { return a + b; }
+};
+struct FMul {
+  __device__ static float apply(float a, float b) { return a * b; }
+};
+struct FMin {
+  __device__ static float apply(float a, float b) { return fminf(a, b); }
+};
+struct FMax {
+  __device__ static float apply(float a, float b) { return fmaxf(a, b); }
+};
+
+// 8 elements per thread via 8-byte loads; fp32 accumulators.
+template <typename Op, bool E5M2>
+__global__ __launch_bounds__(256) void fp8_reduce_kernel(
+    const unsigned char* __restrict__ in, unsigned char* __restrict__ out,
+    long long n, long long nranks) {
+  const long long nvec = n / 8;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       i < nvec; i += stride) {
+    const ulonglong1* base = reinterpret_cast<const ulonglong1*>(in);
+    unsigned long long w = base[i].x;
+    float acc[8];
+    for (int k = 0; k < 8; ++k) {
+      acc[k] = fp8_to_f32<E5M2>((unsigned char)(w >> (8 * k)));
+    }
+    for (long long r = 1; r < nranks; ++r) {
+      unsigned long long wr = base[r * nvec + i].x;  // n % 8 == 0 layout
+      for (int k = 0; k < 8; ++k) {
+        acc[k] = Op::apply(acc[k], fp8_to_f32<E5M2>((unsigned char)(wr >> (8 * k))));
+      }
+    }
+    unsigned long long o = 0;
+    for (int k = 0; k < 8; ++k) {
+      o |= (unsigned long long)f32_to_fp8<E5M2>(acc[k]) << (8 * k);
+    }
+    reinterpret_cast<ulonglong1*>(out)[i].x = o;
+  }
+}
+
+// scalar tail/general path (n not divisible by 8, or unaligned)
+template <typename Op, bool E5M2>
+__global__ __launch_bounds__(256) void fp8_reduce_kernel_scalar(
+    const unsigned char* __restrict__ in, unsigned char* __restrict__ out,
+    long long n, long long nranks) {
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n; i += stride) {
+    float acc = fp8_to_f32<E5M2>(in[i]);
+    for (long long r = 1; r < nranks; ++r) {
+      acc = Op::apply(acc, fp8_to_f32<E5M2>(in[r * n + i]));
+    }
+    out[i] = f32_to_fp8<E5M2>(acc);
+  }
+}
+
+template <typename Op, bool E5M2>
+void launch_fp8_reduce_t(const void* in, void* out, int64_t n, int nranks,
+                         hipStream_t stream) {
+  const bool vec = (n % 8 == 0) &&
+                   (reinterpret_cast<uintptr_t>(in) % 8 == 0) &&
+                   (reinterpret_cast<uintptr_t>(out) % 8 == 0);
+  const long long work = vec ? n / 8 : n;
+  long long blocks = (work + 255) / 256;
+  if (blocks > 4096) blocks = 4096;
+  if (blocks < 1) blocks = 1;
+  if (vec) {
+    hipLaunchKernelGGL((fp8_reduce_kernel<Op, E5M2>), dim3((unsigned)blocks),
+                       dim3(256), 0, stream,
+                       static_cast<const unsigned char*>(in),
+                       static_cast<unsigned char*>(out), (long long)n,
+                       (long long)nranks);
+  } else {
+    hipLaunchKernelGGL((fp8_reduce_kernel_scalar<Op, E5M2>),
+                       dim3((unsigned)blocks), dim3(256), 0, stream,
+                       static_cast<const unsigned char*>(in),
+                       static_cast<unsigned char*>(out), (long long)n,
+                       (long long)nranks);
+  }
+}
+
+} // namespace
+
+void launch_fp8_reduce(const void* in, void* out, int64_t n_elems, int nranks,
+                       int op, bool e5m2, hipStream_t stream) {
+  auto dispatch = [&](auto opv) {
+    using Op = decltype(opv);
+    if (e5m2) {
+      launch_fp8_reduce_t<Op, true>(in, out, n_elems, nranks, stream);
+    } else {
+      launch_fp8_reduce_t<Op, false>(in, out, n_elems, nranks, stream);
+    }
+  };
+  switch (op) {
+    case 0: dispatch(FAdd{}); break;
+    case 1: dispatch(FMul{}); break;
+    case 2: dispatch(FMin{}); break;
+    default: dispatch(FMax{}); break;
   }
 }
 

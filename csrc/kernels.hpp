@@ -46,4 +46,12 @@ void launch_slab_copy(const SlabDesc* descs, int n, hipStream_t stream);
 void launch_bitwise_reduce(const void* in, void* out, int64_t chunk_bytes,
                            int nranks, int op, hipStream_t stream);
 
+// Fused fp8 local reduction with fp32 accumulation: elementwise
+//   out[i] = op_{r<nranks} fp32(in[r*n + i])  quantized to fp8 ONCE.
+// This is the MI355X fp8 allreduce tail (allgather + this kernel): one
+// quantization instead of one per ring hop, fp32 accumulation throughout.
+// op: 0=sum 1=prod 2=min 3=max; e5m2 selects the encoding.
+void launch_fp8_reduce(const void* in, void* out, int64_t n_elems,
+                       int nranks, int op, bool e5m2, hipStream_t stream);
+
 } // namespace m4a

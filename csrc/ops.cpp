@@ -33,6 +33,7 @@
 #include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 
 #include <algorithm>
+#include <cstdlib>
 #include <cstring>
 #include <mutex>
 
@@ -218,9 +219,18 @@ bool native_reduce_dtype(const Transport& tr, at::ScalarType t) {
       case at::kBFloat16:
         return true;
       case at::kFloat8_e4m3fn:
-      case at::kFloat8_e5m2:
-        // native when this RCCL reduces fp8 (runtime probe), else upcast
-        return const_cast<Transport&>(tr).fp8_reduce_supported(t);
+      case at::kFloat8_e5m2: {
+        // Default: the fused allgather + fp32-accumulating CDNA4 kernel
+        // (one quantization; equal wire bytes to an fp32-upcast allreduce
+        // at P=8). Opt into RCCL-native fp8 rings with
+        // MPI4TORCH_AMD_NATIVE_FP8=1 (runtime-probed).
+        static const bool want_native = []() {
+          const char* e = std::getenv("MPI4TORCH_AMD_NATIVE_FP8");
+          return e && e[0] == '1';
+        }();
+        return want_native &&
+               const_cast<Transport&>(tr).fp8_reduce_supported(t);
+      }
       default:
         return false;  // short/bool go through the upcast path
     }
@@ -289,6 +299,23 @@ Tensor allreduce_lowered(Transport& tr, const Tensor& in, int64_t op) {
   if (native_reduce_dtype(tr, in.scalar_type())) {
     auto out = at::empty_like(in);
     tr.allreduce(in, out, (RedOp)op);
+    return out;
+  }
+  const bool fp8 = in.scalar_type() == at::kFloat8_e4m3fn ||
+                   in.scalar_type() == at::kFloat8_e5m2;
+  if (fp8 && tr.is_gpu()) {
+    // fp8 stays fp8 on the wire: allgather the raw bytes, then one fused
+    // CDNA4 reduction with fp32 accumulators and a single quantization
+    // (better numerics than a per-hop-quantizing ring and 4x fewer wire
+    // bytes than the fp32-upcast path per hop).
+    auto staging = at::empty({tr.size() * in.numel()}, in.options());
+    tr.allgather_equal(in, staging);
+    auto out = at::empty_like(in);
+    const int kop = op == kSum ? 0 : (op == kProd ? 1 : (op == kMin ? 2 : 3));
+    launch_fp8_reduce(staging.data_ptr(), out.data_ptr(), in.numel(),
+                      tr.size(), kop,
+                      in.scalar_type() == at::kFloat8_e5m2,
+                      current_gpu_stream(in));
     return out;
   }
   auto up = in.to(upcast_for_reduce(in.scalar_type()));
@@ -1236,6 +1263,20 @@ Tensor debug_pack_roundtrip(const Tensor& input, int64_t axis,
   move_axis_blocks(in, axis, displs, counts, blocks, /*pack=*/true);
   auto out = at::zeros_like(in);
   move_axis_blocks(out, axis, displs, counts, blocks, /*pack=*/false);
+  return out;
+}
+
+Tensor debug_fp8_reduce(const Tensor& stacked, int64_t op) {
+  TORCH_CHECK(stacked.is_cuda(), "debug_fp8_reduce is a GPU kernel test");
+  TORCH_CHECK(stacked.dim() >= 2);
+  TORCH_CHECK(stacked.scalar_type() == at::kFloat8_e4m3fn ||
+              stacked.scalar_type() == at::kFloat8_e5m2);
+  auto in = stacked.contiguous();
+  const int64_t n = in.size(0);
+  auto out = at::empty(in.sizes().slice(1).vec(), in.options());
+  launch_fp8_reduce(in.data_ptr(), out.data_ptr(), out.numel(), (int)n,
+                    (int)op, in.scalar_type() == at::kFloat8_e5m2,
+                    current_gpu_stream(in));
   return out;
 }
 

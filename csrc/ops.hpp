@@ -83,5 +83,6 @@ at::Tensor join_dummies(const at::Tensor& loopthrough,
 at::Tensor debug_pack_roundtrip(const at::Tensor& input, int64_t axis,
                                 std::vector<int64_t> counts);
 at::Tensor debug_bitwise_reduce(const at::Tensor& stacked, int64_t op);
+at::Tensor debug_fp8_reduce(const at::Tensor& stacked, int64_t op);
 
 } // namespace m4a

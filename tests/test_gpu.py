@@ -161,3 +161,39 @@ def test_allreduce_numerics_vs_fp32(world1):
     torch.testing.assert_close(
         r.float(), x32.to(torch.bfloat16).float(), rtol=0, atol=0
     )
+
+
+def test_fp8_reduce_kernel(world1):
+    m = world1
+    # fused fp8 reduce (fp32 accumulation) vs plain PyTorch fp32 reference
+    for enc in (torch.float8_e4m3fn, torch.float8_e5m2):
+        for n in (4096, 4099):  # vector path and scalar-tail path
+            stacked = (torch.randn(5, n, device="cuda") * 0.25).to(enc)
+            ref_in = stacked.float()
+            cases = [
+                (0, lambda a, b: a + b),
+                (1, lambda a, b: a * b),
+                (2, torch.minimum),
+                (3, torch.maximum),
+            ]
+            for op, fn in cases:
+                out = m._C._fp8_reduce(stacked, op)
+                ref = ref_in[0]
+                for i in range(1, 5):
+                    ref = fn(ref, ref_in[i])
+                ref8 = ref.to(enc)
+                torch.cuda.synchronize()
+                assert (out.view(torch.uint8) == ref8.view(torch.uint8)).all(), (
+                    enc, n, op)
+
+
+def test_fp8_allreduce_path(world1):
+    m = world1
+    comm = m.COMM_WORLD
+    # world-1: fused path must still be an exact identity
+    for enc in (torch.float8_e4m3fn, torch.float8_e5m2):
+        t = (torch.randn(1 << 16, device="cuda") * 0.5).to(enc)
+        r = comm.Allreduce(t, m.MPI_SUM)
+        assert (r.view(torch.uint8) == t.view(torch.uint8)).all()
+        r2 = comm.Allreduce(t, m.MPI_MAX)
+        assert (r2.view(torch.uint8) == t.view(torch.uint8)).all()

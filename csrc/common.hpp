@@ -66,6 +66,12 @@ struct Config {
   // hipEvent-based per-collective tracing (the reference has
   // none; we add it as a first-class aux subsystem).
   bool trace_enabled = false;
+  // Collective-desync detector (MPI4TORCH_AMD_DEBUG=1): before every
+  // collective, ranks exchange a hash of (op, shape, dtype, args) over the
+  // host channel and raise on mismatch — turning would-be deadlocks (the
+  // core SPMD hazard, reference doc/basic_usage.rst:184-322) into
+  // immediate, attributed errors.
+  bool debug_collectives = false;
 };
 
 Config& config();

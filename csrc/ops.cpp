@@ -197,6 +197,30 @@ bool is_arith(int64_t op) {
   return op == kSum || op == kProd || op == kMin || op == kMax;
 }
 
+// Collective-desync detector (config().debug_collectives): compare a
+// signature of this collective across all ranks over the host channel and
+// fail with a description instead of deadlocking in RCCL/gloo.
+void debug_check_collective(const std::string& group, const char* opname,
+                            const at::Tensor& t,
+                            std::initializer_list<int64_t> args) {
+  if (!config().debug_collectives || group.empty()) return;
+  size_t h = std::hash<std::string>()(opname);
+  auto mix = [&h](int64_t v) {
+    h ^= std::hash<int64_t>()(v) + 0x9e3779b97f4a7c15ull + (h << 6) + (h >> 2);
+  };
+  mix((int64_t)t.scalar_type());
+  for (auto s : t.sizes()) mix(s);
+  for (auto a : args) mix(a);
+  auto all = host_allgather_int64(group, (int64_t)(h & 0x7fffffffffffffffll));
+  for (size_t r = 1; r < all.size(); ++r) {
+    TORCH_CHECK(all[r] == all[0],
+                "mpi4torch_amd[debug]: collective desync detected at ", opname,
+                " — rank ", r, " issued a different op/shape/dtype/arguments "
+                "than rank 0. This would deadlock without "
+                "MPI4TORCH_AMD_DEBUG=1.");
+  }
+}
+
 void check_op(int64_t op) {
   TORCH_CHECK(op >= kMax && op <= kMaxLoc, "invalid reduction op ", op);
   TORCH_CHECK(op != kMinLoc && op != kMaxLoc,
@@ -417,6 +441,7 @@ Tensor Communicator::Allreduce(const Tensor& input, int64_t op) {
     DeviceStager stager(input);
     auto in = stager.to_comm(input).contiguous().variable_data();
     auto& tr = tr_for(in);
+    debug_check_collective(group_name_, "Allreduce", in, {op});
     return stager.from_comm(allreduce_lowered(tr, in, op));
   }();
   attach_history(result, grad_fn);
@@ -455,6 +480,7 @@ Tensor Communicator::Bcast_(const Tensor& input, int64_t root) {
     DeviceStager stager(input);
     auto t = stager.to_comm(input).contiguous().variable_data();
     auto& tr = tr_for(t);
+    debug_check_collective(group_name_, "Bcast_", t, {root});
     if (tr.size() > 1) tr.broadcast(t, (int)root);
     return stager.from_comm(std::move(t));
   }();
@@ -500,6 +526,7 @@ Tensor Communicator::Reduce_(const Tensor& input, int64_t op, int64_t root) {
     DeviceStager stager(input);
     auto t = stager.to_comm(input).contiguous().variable_data();
     auto& tr = tr_for(t);
+    debug_check_collective(group_name_, "Reduce_", t, {op, root});
     if (tr.size() > 1) {
       if (is_arith(op) && native_reduce_dtype(tr, t.scalar_type())) {
         tr.reduce(t, (RedOp)op, (int)root);
@@ -659,6 +686,7 @@ Tensor Communicator::Gather(const Tensor& input, int64_t gatheraxis,
     auto in = stager.to_comm(input).contiguous().variable_data();
     auto& tr = tr_for(in);
     if (tr.size() == 1) return stager.from_comm(in.clone());
+    debug_check_collective(group_name_, "Gather", at::empty({0}), {gatheraxis, root});
     const auto g = axis_geom(in, gatheraxis);
     auto counts = host_allgather_int64(group_name_, g.axis);
     auto displs = prefix_displs(counts);

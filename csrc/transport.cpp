@@ -18,6 +18,7 @@
 #include <rccl/rccl.h>
 
 #include <atomic>
+#include <cstdlib>
 #include <cstring>
 #include <deque>
 #include <map>
@@ -27,7 +28,12 @@
 namespace m4a {
 
 Config& config() {
-  static Config cfg;
+  static Config cfg = []() {
+    Config c;
+    const char* dbg = std::getenv("MPI4TORCH_AMD_DEBUG");
+    c.debug_collectives = dbg && dbg[0] == '1';
+    return c;
+  }();
   return cfg;
 }
 

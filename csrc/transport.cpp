@@ -531,7 +531,7 @@ struct RcclTransport final : Transport {
     M4A_NCCL_CHECK(ncclAllReduce(in.data_ptr(), out.data_ptr(), in.numel(),
                                  nccl_reduce_dtype(in.scalar_type()),
                                  nccl_red_op(op), comm(Channel::Coll),
-                                 stream(Channel::Coll)));
+                                 hop.stream()));
   }
   void broadcast(at::Tensor& t, int root) override {
     std::lock_guard<std::mutex> g(mu_);
@@ -539,7 +539,7 @@ struct RcclTransport final : Transport {
     // byte-typed: broadcast moves bytes, dtype-agnostic
     M4A_NCCL_CHECK(ncclBroadcast(t.data_ptr(), t.data_ptr(), nbytes(t),
                                  ncclUint8, root, comm(Channel::Coll),
-                                 stream(Channel::Coll)));
+                                 hop.stream()));
   }
   void reduce(at::Tensor& t, RedOp op, int root) override {
     std::lock_guard<std::mutex> g(mu_);
@@ -547,14 +547,14 @@ struct RcclTransport final : Transport {
     M4A_NCCL_CHECK(ncclReduce(t.data_ptr(), t.data_ptr(), t.numel(),
                               nccl_reduce_dtype(t.scalar_type()),
                               nccl_red_op(op), root, comm(Channel::Coll),
-                              stream(Channel::Coll)));
+                              hop.stream()));
   }
   void allgather_equal(const at::Tensor& in, at::Tensor& out) override {
     std::lock_guard<std::mutex> g(mu_);
     Hop hop(*this, Channel::Coll, {in, out});
     M4A_NCCL_CHECK(ncclAllGather(in.data_ptr(), out.data_ptr(), nbytes(in),
                                  ncclUint8, comm(Channel::Coll),
-                                 stream(Channel::Coll)));
+                                 hop.stream()));
   }
   void reduce_scatter_equal(const at::Tensor& in, at::Tensor& out,
                             RedOp op) override {
@@ -564,7 +564,7 @@ struct RcclTransport final : Transport {
                                      out.numel(),
                                      nccl_reduce_dtype(out.scalar_type()),
                                      nccl_red_op(op), comm(Channel::Coll),
-                                     stream(Channel::Coll)));
+                                     hop.stream()));
   }
   void exchange(const std::vector<at::Tensor>& sendbufs,
                 const std::vector<int>& speers,
@@ -589,7 +589,7 @@ struct RcclTransport final : Transport {
       if (nbytes(sendbufs[i]) == 0) continue;
       M4A_NCCL_CHECK(ncclSend(sendbufs[i].data_ptr(), nbytes(sendbufs[i]),
                               ncclUint8, speers[i], comm(Channel::Coll),
-                              stream(Channel::Coll)));
+                              hop.stream()));
     }
     for (size_t j = 0; j < recvbufs.size(); ++j) {
       if (rpeers[j] == me) {
@@ -599,7 +599,7 @@ struct RcclTransport final : Transport {
       if (nbytes(recvbufs[j]) == 0) continue;
       M4A_NCCL_CHECK(ncclRecv(recvbufs[j].data_ptr(), nbytes(recvbufs[j]),
                               ncclUint8, rpeers[j], comm(Channel::Coll),
-                              stream(Channel::Coll)));
+                              hop.stream()));
     }
     M4A_NCCL_CHECK(ncclGroupEnd());
     TORCH_CHECK(self_sends.size() == self_recvs.size(),
@@ -608,14 +608,14 @@ struct RcclTransport final : Transport {
       // device copy on the collective stream, inside the hop bracket
       M4A_HIP_CHECK(hipMemcpyAsync(
           self_recvs[i]->data_ptr(), self_sends[i]->data_ptr(),
-          nbytes(*self_sends[i]), hipMemcpyDeviceToDevice,
-          stream(Channel::Coll)));
+          nbytes(*self_sends[i]), hipMemcpyDeviceToDevice, hop.stream()));
     }
   }
   uint64_t isend(const at::Tensor& buf, int peer, int tag,
                  Channel ch) override {
     if (peer == rank_) return self_matcher().isend(buf, tag, ch);
     std::lock_guard<std::mutex> g(mu_);
+    check_not_capturing("Isend");
     enter_side(ch, {buf});
     M4A_NCCL_CHECK(ncclSend(buf.data_ptr(), nbytes(buf), ncclUint8, peer,
                             comm(ch), stream(ch)));
@@ -624,6 +624,7 @@ struct RcclTransport final : Transport {
   uint64_t irecv(at::Tensor& buf, int peer, int tag, Channel ch) override {
     if (peer == rank_) return self_matcher().irecv(buf, tag, ch);
     std::lock_guard<std::mutex> g(mu_);
+    check_not_capturing("Irecv");
     enter_side(ch, {buf});
     M4A_NCCL_CHECK(ncclRecv(buf.data_ptr(), nbytes(buf), ncclUint8, peer,
                             comm(ch), stream(ch)));
@@ -656,6 +657,7 @@ struct RcclTransport final : Transport {
     // only waits when the returned request is waited — this is what lets
     // gradient-bucket allreduce overlap the rest of backward.
     std::lock_guard<std::mutex> g(mu_);
+    check_not_capturing("Iallreduce");
     enter_side(Channel::Coll, {in, out});
     M4A_NCCL_CHECK(ncclAllReduce(in.data_ptr(), out.data_ptr(), in.numel(),
                                  nccl_reduce_dtype(in.scalar_type()),
@@ -671,10 +673,24 @@ struct RcclTransport final : Transport {
   ncclComm_t comm(Channel ch) { return comms_[(int)ch]; }
   hipStream_t stream(Channel ch) { return streams_[(int)ch].stream(); }
 
+  static bool stream_capturing(hipStream_t s) {
+    hipStreamCaptureStatus st = hipStreamCaptureStatusNone;
+    (void)hipStreamIsCapturing(s, &st);
+    return st != hipStreamCaptureStatusNone;
+  }
+
   // Failure detection (the reference has none beyond ierr
   // checks): surface asynchronous RCCL errors (peer crash, xGMI fault,
   // aborted communicator) as exceptions at the next collective instead of
   // hanging the job.
+  void check_not_capturing(const char* what) {
+    auto cur = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(device_);
+    TORCH_CHECK(!stream_capturing(cur.stream()),
+                "mpi4torch_amd: ", what,
+                " inside hipGraph capture is not supported (its Wait may "
+                "fall outside the graph); capture blocking collectives only");
+  }
+
   void check_async_errors() {
     for (int i = 0; i < 3; ++i) {
       if (!comms_[i]) continue;
@@ -725,14 +741,30 @@ struct RcclTransport final : Transport {
     return requests().add(std::move(r));
   }
 
+  // Stream bracket for one collective. Under hipGraph capture the op runs
+  // directly on the capturing stream (events/recordStream are skipped: the
+  // graph serializes ordering and graph memory pools own lifetimes) so
+  // collectives are capturable like ProcessGroupNCCL's.
   struct Hop {
     Hop(RcclTransport& t, Channel ch, const std::vector<at::Tensor>& ts)
         : t_(t), ch_(ch) {
-      t_.enter_side(ch_, ts);
+      auto cur = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(t_.device_);
+      capturing_ = stream_capturing(cur.stream());
+      if (capturing_) {
+        use_stream_ = cur.stream();
+      } else {
+        t_.enter_side(ch_, ts);
+        use_stream_ = t_.stream(ch_);
+      }
     }
-    ~Hop() { t_.exit_side(ch_); }
+    ~Hop() {
+      if (!capturing_) t_.exit_side(ch_);
+    }
+    hipStream_t stream() const { return use_stream_; }
     RcclTransport& t_;
     Channel ch_;
+    hipStream_t use_stream_;
+    bool capturing_ = false;
   };
 
   int device_;

@@ -197,3 +197,27 @@ def test_fp8_allreduce_path(world1):
         assert (r.view(torch.uint8) == t.view(torch.uint8)).all()
         r2 = comm.Allreduce(t, m.MPI_MAX)
         assert (r2.view(torch.uint8) == t.view(torch.uint8)).all()
+
+
+def test_hipgraph_capture(world1):
+    m = world1
+    comm = m.COMM_WORLD
+    static_in = torch.rand(1 << 18, device="cuda")
+    x0 = static_in.clone()
+    # warmup on a side stream (torch graph-capture protocol)
+    s = torch.cuda.Stream()
+    s.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(s):
+        for _ in range(3):
+            comm.Allreduce(static_in, m.MPI_SUM)
+    torch.cuda.current_stream().wait_stream(s)
+
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        static_out = comm.Allreduce(static_in, m.MPI_SUM)
+        static_out2 = comm.Gather(static_out, 0, 0)
+    for i in range(3):
+        static_in.copy_(x0 * (i + 1))
+        g.replay()
+        torch.cuda.synchronize()
+        torch.testing.assert_close(static_out2, x0 * (i + 1))

@@ -290,7 +290,7 @@ at::ScalarType upcast_for_reduce(at::ScalarType t) {
 
 // Elementwise allreduce with full op/dtype lowering. `in` contiguous.
 Tensor allreduce_lowered(Transport& tr, const Tensor& in, int64_t op) {
-  if (tr.size() == 1) return in.clone();
+  if (tr.size() == 1 || in.numel() == 0) return in.clone();
   if (is_logical(op)) {
     // land/lor/lxor lower to min/max/sum over 0/1 indicators; valid on any
     // dtype and on both transports (RCCL has no logical ops).
@@ -481,7 +481,7 @@ Tensor Communicator::Bcast_(const Tensor& input, int64_t root) {
     auto t = stager.to_comm(input).contiguous().variable_data();
     auto& tr = tr_for(t);
     debug_check_collective(group_name_, "Bcast_", t, {root});
-    if (tr.size() > 1) tr.broadcast(t, (int)root);
+    if (tr.size() > 1 && t.numel() > 0) tr.broadcast(t, (int)root);
     return stager.from_comm(std::move(t));
   }();
   attach_history(result, grad_fn);
@@ -527,7 +527,7 @@ Tensor Communicator::Reduce_(const Tensor& input, int64_t op, int64_t root) {
     auto t = stager.to_comm(input).contiguous().variable_data();
     auto& tr = tr_for(t);
     debug_check_collective(group_name_, "Reduce_", t, {op, root});
-    if (tr.size() > 1) {
+    if (tr.size() > 1 && t.numel() > 0) {
       if (is_arith(op) && native_reduce_dtype(tr, t.scalar_type())) {
         tr.reduce(t, (RedOp)op, (int)root);
       } else {

@@ -102,7 +102,12 @@ class DistributedDataParallel(torch.nn.Module):
 
     def finish_gradient_sync(self):
         """Wait all in-flight bucket allreduces and scatter results back
-        into .grad. Call between backward() and optimizer.step()."""
+        into .grad. Call between backward() and optimizer.step().
+
+        Limitation: every parameter of the wrapped module must receive a
+        gradient each backward (no unused-parameter detection yet) — a
+        bucket whose members only partially produced gradients is skipped,
+        and those gradients would stay rank-local."""
         if self.comm.size == 1:
             return
         scale = 1.0 / self.comm.size if self.average else 1.0

@@ -221,3 +221,36 @@ def test_hipgraph_capture(world1):
         g.replay()
         torch.cuda.synchronize()
         torch.testing.assert_close(static_out2, x0 * (i + 1))
+
+
+def test_soak_no_leaks(world1):
+    m = world1
+    comm = m.COMM_WORLD
+    torch.cuda.synchronize()
+    torch.cuda.empty_cache()
+    base = torch.cuda.memory_allocated()
+    x = torch.rand(2, 64, 8, device="cuda")
+    for i in range(200):
+        t = torch.rand(1 << 14, device="cuda").requires_grad_()
+        r = comm.Allreduce(t, m.MPI_SUM)
+        r.backward(torch.ones_like(r))
+        comm.Gather(x, 1, 0)
+        comm.Alltoall(x, 0, 1, 64)
+        h = comm.Isend(x, 0, i)
+        h2 = comm.Irecv(torch.empty_like(x), 0, i)
+        comm.Wait(h2)
+        comm.Wait(h)
+        hh = comm.Iallreduce(x.reshape(-1), m.MPI_SUM)
+        comm.Wait(hh)
+    torch.cuda.synchronize()
+    grown = torch.cuda.memory_allocated() - base
+    # event pool + request table must not accumulate device memory
+    assert grown < 32 * 1024 * 1024, f"leaked {grown} bytes over 200 iters"
+
+
+def test_zero_numel_ops(world1):
+    m = world1
+    comm = m.COMM_WORLD
+    e = torch.empty(0, device="cuda")
+    assert comm.Allreduce(e, m.MPI_SUM).numel() == 0
+    assert comm.Bcast_(e.clone(), 0).numel() == 0

@@ -740,6 +740,8 @@ Tensor Communicator::Scatter(const Tensor& input, int64_t scatteraxis,
       return stager.from_comm(in.clone());
     }
     const int me = tr.rank();
+    debug_check_collective(group_name_, "Scatter", at::empty({0}),
+                           {scatteraxis, root});
     // root broadcasts [ndim, sizes...]: the shape contract for non-root
     // ranks whose input tensor is a placeholder (reference :788-796)
     std::vector<int64_t> meta;
@@ -820,6 +822,8 @@ Tensor Communicator::Allgather(const Tensor& input, int64_t gatheraxis) {
       }
       return stager.from_comm(in.clone());
     }
+    debug_check_collective(group_name_, "Allgather", at::empty({0}),
+                           {gatheraxis});
     const auto g = axis_geom(in, gatheraxis);
     auto counts = host_allgather_int64(group_name_, g.axis);
     auto displs = prefix_displs(counts);
@@ -972,6 +976,8 @@ Tensor Communicator::Alltoall(const Tensor& input, int64_t gatheraxis,
     }
     const int me = tr.rank();
     const int P = tr.size();
+    debug_check_collective(group_name_, "Alltoall", at::empty({0}),
+                           {gatheraxis, scatteraxis});
 
     // What each rank wants along the scatter axis:
     auto scounts = host_allgather_int64(group_name_, numelem);

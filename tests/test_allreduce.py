@@ -143,3 +143,22 @@ def test_allreduce_dtypes_ws2():
 
 def test_allreduce_errors_ws2():
     run_spmd(2, _errors_worker)
+
+
+def _double_backward_worker(rank, world):
+    import mpi4torch_amd as m
+
+    comm = m.COMM_WORLD
+    # second-order gradients through the self-adjoint Allreduce:
+    # y = allreduce(x); L = sum(y^2); dL/dx = allreduce(2*allreduce(x));
+    # d/dx sum(dL/dx * v) for v=ones -> allreduce(allreduce(2*ones)) = 2*P^2
+    x = torch.rand(6, dtype=torch.double, requires_grad=True)
+    y = comm.Allreduce(x, m.MPI_SUM)
+    loss = (y ** 2).sum()
+    (g,) = torch.autograd.grad(loss, x, create_graph=True)
+    (gg,) = torch.autograd.grad(g.sum(), x)
+    assert torch.allclose(gg, torch.full_like(x, 2.0 * world * world)), gg
+
+
+def test_double_backward_ws2():
+    run_spmd(2, _double_backward_worker)

@@ -89,6 +89,10 @@ def test_pack_kernel_roundtrip(world1):
         ((64, 33), 0, [10, 20, 34], torch.float32),
         ((2, 9, 3), 1, [9], torch.float8_e4m3fn),    # 1-byte dtype
         ((8, 1024, 256), 1, [256, 512, 256], torch.bfloat16),  # big, vector path
+        # odd phases: exercise the byte-window (shift) path
+        ((8, 65537, 511), 1, [8192, 30000, 27345], torch.bfloat16),
+        ((3, 1001, 7), 1, [137, 500, 364], torch.float8_e4m3fn),
+        ((1, 999983), 0, [1, 2, 999980], torch.bfloat16),
     ]
     for shape, axis, counts, dtype in cases:
         x = (torch.randn(shape, device="cuda", dtype=torch.float32)).to(dtype)

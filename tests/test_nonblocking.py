@@ -120,3 +120,33 @@ def test_forward_values_ws5():
 
 def test_bifurcation_detection_ws2():
     run_spmd(2, _bifurcation_worker)
+
+
+def _channel_independence_worker(rank, world):
+    import mpi4torch_amd as m
+
+    comm = m.COMM_WORLD
+    # p2p and collectives are posted in DIFFERENT orders on the two ranks —
+    # legal under MPI semantics because a posted Isend does not block the
+    # sender. The dedicated p2p channel (transport.hpp) preserves this;
+    # a single-ordered transport would deadlock here.
+    x = torch.full((1000,), float(rank))
+    y = torch.ones(10)
+    if rank == 0:
+        h = comm.Isend(x, 1, 9)
+        ar = comm.Allreduce(y, m.MPI_SUM)
+        comm.Wait(h)
+    else:
+        ar = comm.Allreduce(y, m.MPI_SUM)
+        if rank == 1:
+            got = comm.Recv(torch.empty(1000), 0, 9)
+            assert (got == 0.0).all()
+    assert (ar == world).all()
+
+
+def test_channel_independence_ws2():
+    run_spmd(2, _channel_independence_worker)
+
+
+def test_channel_independence_ws5():
+    run_spmd(5, _channel_independence_worker)

@@ -165,11 +165,13 @@ __global__ __launch_bounds__(256) void slab_copy_shift_kernel(
     const long long body = (a.row_b - head) / 16;
     const char* sbase = srow + head;
     const int s = (int)((uintptr_t)sbase & 15);
-    // with a nonzero shift the last vector chunk would read up to 15 bytes
-    // past the row: hand it to the bytewise tail instead
+    // Nonzero shift: the last vector chunk would read up to 15 bytes past
+    // the row, and chunk 0's aligned base (sbase - s) sits up to 15 bytes
+    // BEFORE the row when s > head — both go to the bytewise worker.
     const long long bodyv = (s == 0) ? body : (body > 0 ? body - 1 : 0);
-    if (c < bodyv) {
-      const char* sal = sbase - s + 16 * c;  // 16-aligned
+    const long long vs = (s == 0 || s <= head) ? 0 : 1;  // first vector chunk
+    if (c >= vs && c < bodyv) {
+      const char* sal = sbase - s + 16 * c;  // 16-aligned, in-bounds
       if (s == 0) {
         *reinterpret_cast<uint4*>(drow + head + 16 * c) =
             *reinterpret_cast<const uint4*>(sal);
@@ -180,9 +182,15 @@ __global__ __launch_bounds__(256) void slab_copy_shift_kernel(
             byte_window(lo, hi, s);
       }
     } else if (c == bodyv) {
-      for (long long k = 0; k < head; ++k) drow[k] = srow[k];
-      for (long long k = head + bodyv * 16; k < a.row_b; ++k) {
-        drow[k] = srow[k];
+      // bytewise worker: head (+ the excluded chunk 0, if any) and tail
+      if (bodyv <= vs) {
+        for (long long k = 0; k < a.row_b; ++k) drow[k] = srow[k];
+      } else {
+        const long long lo_end = head + vs * 16;
+        for (long long k = 0; k < lo_end; ++k) drow[k] = srow[k];
+        for (long long k = head + bodyv * 16; k < a.row_b; ++k) {
+          drow[k] = srow[k];
+        }
       }
     }
   }

@@ -92,7 +92,7 @@ def test_pack_kernel_roundtrip(world1):
         # odd phases: exercise the byte-window (shift) path
         ((8, 65537, 511), 1, [8192, 30000, 27345], torch.bfloat16),
         ((3, 1001, 7), 1, [137, 500, 364], torch.float8_e4m3fn),
-        ((1, 999983), 0, [1, 2, 999980], torch.bfloat16),
+        ((1, 999983), 1, [1, 2, 999980], torch.bfloat16),
     ]
     for shape, axis, counts, dtype in cases:
         x = (torch.randn(shape, device="cuda", dtype=torch.float32)).to(dtype)

@@ -72,6 +72,10 @@ struct Config {
   // core SPMD hazard, reference doc/basic_usage.rst:184-322) into
   // immediate, attributed errors.
   bool debug_collectives = false;
+  // MPI4TORCH_AMD_FORCE_FULL_PATH=1: disable world-size-1 fast paths so the
+  // full pack -> exchange -> unpack machinery runs even alone (kernel
+  // benchmarking / native-path verification on a single GPU).
+  bool force_full_path = false;
 };
 
 Config& config();

@@ -90,6 +90,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("host_staging_forced",
         []() { return m4a::config().force_host_staging; });
 
+  // Disable world-size-1 fast paths (kernel benchmarking / native-path
+  // verification on a single GPU); also via MPI4TORCH_AMD_FORCE_FULL_PATH=1.
+  m.def("force_full_path",
+        [](bool enabled) { m4a::config().force_full_path = enabled; });
+
   m.def("_pack_roundtrip", &m4a::debug_pack_roundtrip);
   m.def("_bitwise_reduce", &m4a::debug_bitwise_reduce);
   m.def("_fp8_reduce", &m4a::debug_fp8_reduce);

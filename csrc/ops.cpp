@@ -289,8 +289,17 @@ at::ScalarType upcast_for_reduce(at::ScalarType t) {
 }
 
 // Elementwise allreduce with full op/dtype lowering. `in` contiguous.
+bool w1_shortcut(const Transport& tr) {
+  return tr.size() == 1 && !config().force_full_path;
+}
+
 Tensor allreduce_lowered(Transport& tr, const Tensor& in, int64_t op) {
-  if (tr.size() == 1 || in.numel() == 0) return in.clone();
+  if (in.numel() == 0) return in.clone();
+  if (tr.size() == 1 &&
+      (!config().force_full_path || !tr.is_gpu() || !is_arith(op) ||
+       !native_reduce_dtype(tr, in.scalar_type()))) {
+    return in.clone();
+  }
   if (is_logical(op)) {
     // land/lor/lxor lower to min/max/sum over 0/1 indicators; valid on any
     // dtype and on both transports (RCCL has no logical ops).
@@ -685,7 +694,7 @@ Tensor Communicator::Gather(const Tensor& input, int64_t gatheraxis,
     DeviceStager stager(input);
     auto in = stager.to_comm(input).contiguous().variable_data();
     auto& tr = tr_for(in);
-    if (tr.size() == 1) return stager.from_comm(in.clone());
+    if (w1_shortcut(tr)) return stager.from_comm(in.clone());
     debug_check_collective(group_name_, "Gather", at::empty({0}), {gatheraxis, root});
     const auto g = axis_geom(in, gatheraxis);
     auto counts = host_allgather_int64(group_name_, g.axis);
@@ -734,7 +743,7 @@ Tensor Communicator::Scatter(const Tensor& input, int64_t scatteraxis,
     DeviceStager stager(input);
     auto in = stager.to_comm(input).contiguous().variable_data();
     auto& tr = tr_for(in);
-    if (tr.size() == 1) {
+    if (w1_shortcut(tr)) {
       TORCH_CHECK(numelem == in.size(at::maybe_wrap_dim(scatteraxis, in.dim())),
                   "Scatter: numelem must equal the axis size at world size 1");
       return stager.from_comm(in.clone());
@@ -813,7 +822,7 @@ Tensor Communicator::Allgather(const Tensor& input, int64_t gatheraxis) {
     DeviceStager stager(input);
     auto in = stager.to_comm(input).contiguous().variable_data();
     auto& tr = tr_for(in);
-    if (tr.size() == 1) {
+    if (w1_shortcut(tr)) {
       if (needs_grad) {
         grad_fn = make_node<AllgatherBackward>(
             c10::intrusive_ptr<Communicator>::unsafe_reclaim_from_nonowning(
@@ -968,7 +977,7 @@ Tensor Communicator::Alltoall(const Tensor& input, int64_t gatheraxis,
     DeviceStager stager(input);
     auto in = stager.to_comm(input).contiguous().variable_data();
     auto& tr = tr_for(in);
-    if (tr.size() == 1) {
+    if (w1_shortcut(tr)) {
       TORCH_CHECK(numelem == in.size(scatteraxis),
                   "Alltoall: numelem must equal the scatter-axis size at "
                   "world size 1");

@@ -32,6 +32,8 @@ Config& config() {
     Config c;
     const char* dbg = std::getenv("MPI4TORCH_AMD_DEBUG");
     c.debug_collectives = dbg && dbg[0] == '1';
+    const char* ffp = std::getenv("MPI4TORCH_AMD_FORCE_FULL_PATH");
+    c.force_full_path = ffp && ffp[0] == '1';
     return c;
   }();
   return cfg;

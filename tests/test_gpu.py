@@ -258,3 +258,28 @@ def test_zero_numel_ops(world1):
     e = torch.empty(0, device="cuda")
     assert comm.Allreduce(e, m.MPI_SUM).numel() == 0
     assert comm.Bcast_(e.clone(), 0).numel() == 0
+
+
+def test_full_path_axis_collectives(world1):
+    m = world1
+    comm = m.COMM_WORLD
+    # run the real pack -> exchange(self) -> unpack pipeline at world 1
+    m._C.force_full_path(True)
+    try:
+        x = torch.rand(2, 5, 12, 3, device="cuda", dtype=torch.bfloat16)
+        torch.testing.assert_close(comm.Gather(x, 2, 0), x)
+        torch.testing.assert_close(comm.Allgather(x, 2), x)
+        torch.testing.assert_close(comm.Scatter(x, 2, 12, 0), x)
+        torch.testing.assert_close(comm.Alltoall(x, 1, 2, 12), x)
+        torch.testing.assert_close(comm.Alltoall(x, 2, 2, 12), x)  # same axis
+        t = torch.rand(1 << 16, device="cuda").requires_grad_()
+        r = comm.Allreduce(t, m.MPI_SUM)  # true ncclAllReduce at world 1
+        torch.testing.assert_close(r, t.detach())
+        r.backward(torch.ones_like(r))
+        torch.testing.assert_close(t.grad, torch.ones_like(t))
+        # autograd through the full gather/scatter pipeline
+        y = torch.rand(2, 7, 3, device="cuda", dtype=torch.float32).requires_grad_()
+        comm.Gather(y, 1, 0).sum().backward()
+        torch.testing.assert_close(y.grad, torch.ones_like(y))
+    finally:
+        m._C.force_full_path(False)

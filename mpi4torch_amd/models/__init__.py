@@ -6,5 +6,6 @@ doc/examples.rst). These modules are those patterns as reusable code.
 """
 
 from .linreg import DistributedLinReg
+from .transformer import UlyssesTransformerBlock
 
-__all__ = ["DistributedLinReg"]
+__all__ = ["DistributedLinReg", "UlyssesTransformerBlock"]

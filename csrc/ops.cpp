@@ -664,7 +664,10 @@ std::vector<Tensor> make_blocks(const Tensor& like, int64_t before,
   for (size_t i = 0; i < counts.size(); ++i) {
     const int64_t n = before * counts[i] * after;
     if (direct) {
-      blocks[i] = flat.narrow(0, displs[i] * after, n);
+      // zero-length slices may carry an offset past the end (empty overlap
+      // intervals in same-axis Alltoall): clamp to a valid empty view
+      blocks[i] = n == 0 ? flat.narrow(0, 0, 0)
+                         : flat.narrow(0, displs[i] * after, n);
     } else {
       blocks[i] = at::empty({n}, like.options());
     }

@@ -76,6 +76,11 @@ struct Config {
   // full pack -> exchange -> unpack machinery runs even alone (kernel
   // benchmarking / native-path verification on a single GPU).
   bool force_full_path = false;
+  // MPI4TORCH_AMD_TIMEOUT_S: bound every blocking CPU-transport wait; a
+  // desynchronized peer then raises instead of hanging forever. 0 = wait
+  // indefinitely (default). (GPU-side progress is stream-ordered; use the
+  // desync detector for pre-enqueue divergence.)
+  int64_t op_timeout_ms = 0;
 };
 
 Config& config();

@@ -34,6 +34,9 @@ Config& config() {
     c.debug_collectives = dbg && dbg[0] == '1';
     const char* ffp = std::getenv("MPI4TORCH_AMD_FORCE_FULL_PATH");
     c.force_full_path = ffp && ffp[0] == '1';
+    if (const char* to = std::getenv("MPI4TORCH_AMD_TIMEOUT_S")) {
+      c.op_timeout_ms = (int64_t)(std::atof(to) * 1000.0);
+    }
     return c;
   }();
   return cfg;
@@ -274,6 +277,18 @@ c10d::ReduceOp c10d_red_op(RedOp op) {
   }
 }
 
+void wait_work(const c10::intrusive_ptr<c10d::Work>& w) {
+  const int64_t ms = config().op_timeout_ms;
+  if (ms > 0) {
+    TORCH_CHECK(w->wait(std::chrono::milliseconds(ms)),
+                "mpi4torch_amd: collective timed out after ", ms,
+                " ms (MPI4TORCH_AMD_TIMEOUT_S) — a peer likely diverged; "
+                "run with MPI4TORCH_AMD_DEBUG=1 to locate the desync");
+  } else {
+    w->wait();
+  }
+}
+
 c10::intrusive_ptr<c10d::Backend> gloo_backend(const std::string& group_name) {
   auto pg = c10d::resolve_process_group(group_name);
   TORCH_CHECK(pg, "mpi4torch_amd: process group '", group_name,
@@ -357,14 +372,14 @@ struct C10dTransport final : Transport {
     std::vector<at::Tensor> ts{out};
     c10d::AllreduceOptions opts;
     opts.reduceOp = c10d_red_op(op);
-    backend_->allreduce(ts, opts)->wait();
+    wait_work(backend_->allreduce(ts, opts));
   }
   void broadcast(at::Tensor& t, int root) override {
     std::lock_guard<std::mutex> g(mu_);
     std::vector<at::Tensor> ts{t};
     c10d::BroadcastOptions opts;
     opts.rootRank = root;
-    backend_->broadcast(ts, opts)->wait();
+    wait_work(backend_->broadcast(ts, opts));
   }
   void reduce(at::Tensor& t, RedOp op, int root) override {
     std::lock_guard<std::mutex> g(mu_);
@@ -372,7 +387,7 @@ struct C10dTransport final : Transport {
     c10d::ReduceOptions opts;
     opts.reduceOp = c10d_red_op(op);
     opts.rootRank = root;
-    backend_->reduce(ts, opts)->wait();
+    wait_work(backend_->reduce(ts, opts));
   }
   void allgather_equal(const at::Tensor& in, at::Tensor& out) override {
     std::lock_guard<std::mutex> g(mu_);
@@ -381,7 +396,7 @@ struct C10dTransport final : Transport {
     for (auto& c : chunks) outs.push_back(c.view_as(in));
     std::vector<std::vector<at::Tensor>> outputs{outs};
     std::vector<at::Tensor> inputs{const_cast<at::Tensor&>(in)};
-    backend_->allgather(outputs, inputs)->wait();
+    wait_work(backend_->allgather(outputs, inputs));
   }
   void reduce_scatter_equal(const at::Tensor& in, at::Tensor& out,
                             RedOp op) override {
@@ -392,7 +407,7 @@ struct C10dTransport final : Transport {
       std::vector<at::Tensor> ts{tmp};
       c10d::AllreduceOptions opts;
       opts.reduceOp = c10d_red_op(op);
-      backend_->allreduce(ts, opts)->wait();
+      wait_work(backend_->allreduce(ts, opts));
     }
     out.copy_(tmp.view({size(), -1})[rank()].view_as(out));
   }
@@ -432,7 +447,7 @@ struct C10dTransport final : Transport {
       // they always match in element count
       self_recvs[i]->view({-1}).copy_(self_sends[i]->reshape({-1}));
     }
-    for (auto& w : works) w->wait();
+    for (auto& w : works) wait_work(w);
   }
   uint64_t isend(const at::Tensor& buf, int peer, int tag,
                  Channel ch) override {
@@ -789,7 +804,7 @@ void wait_request(uint64_t id) {
     M4A_HIP_CHECK(hipStreamWaitEvent(cur.stream(), r.event, 0));
     EventPool::forDevice(r.device).release(r.event);
   } else if (r.work) {
-    r.work->wait();
+    wait_work(r.work);
   }
 }
 

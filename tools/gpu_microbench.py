@@ -63,6 +63,29 @@ def main():
     nbytes = stacked.numel() * 4 + stacked[0].numel() * 4
     print(f"8x256MiB int32 AND: {t*1e3:.3f} ms  {nbytes/t/1e12:.2f} TB/s")
 
+    print("== fused fp8 reduce kernel (fp32 accumulation) ==")
+    stacked8 = (torch.randn(8, 1 << 27, device="cuda") * 0.2).to(
+        torch.float8_e4m3fn)  # 8 x 128 MiB
+    t = timed(lambda: m._C._fp8_reduce(stacked8, 0))
+    nbytes = stacked8.numel() + stacked8[0].numel()
+    print(f"8x128MiB fp8 SUM: {t*1e3:.3f} ms  {nbytes/t/1e12:.2f} TB/s")
+
+    print("== full-path axis collectives (world 1, pack+exchange+unpack) ==")
+    m._C.force_full_path(True)
+    try:
+        comm1 = m.COMM_WORLD
+        xb = torch.randn(8, 65536, 256, device="cuda", dtype=torch.float32).to(
+            torch.bfloat16)  # 256 MiB, middle axis
+        t = timed(lambda: comm1.Allgather(xb, 1))
+        nbytes = xb.numel() * 2
+        print(f"Allgather mid-axis 256MiB: {t*1e3:.3f} ms "
+              f"{nbytes/t/1e9:.1f} GB/s payload")
+        t = timed(lambda: comm1.Alltoall(xb, 1, 1, 65536))
+        print(f"Alltoall same-axis 256MiB: {t*1e3:.3f} ms "
+              f"{nbytes/t/1e9:.1f} GB/s payload")
+    finally:
+        m._C.force_full_path(False)
+
     print("== world-1 RCCL allreduce (copy bound) ==")
     comm = m.COMM_WORLD
     for mib in (64, 1024):

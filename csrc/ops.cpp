@@ -35,7 +35,9 @@
 #include <algorithm>
 #include <cstdlib>
 #include <cstring>
+#include <map>
 #include <mutex>
+#include <tuple>
 
 namespace m4a {
 
@@ -182,6 +184,13 @@ void move_axis_blocks(const Tensor& full, int64_t axis,
     }
   }
 }
+
+// NOTE on the count exchanges below: a "static shapes" cache keyed by a
+// rank's OWN contribution was tried and removed — own-value repetition
+// does not imply the other ranks repeat, so a partial cache hit splits
+// the gloo exchange and deadlocks. The ~0.1 ms host round-trip per axis
+// collective stays; it is off the critical path for multi-megabyte
+// payloads.
 
 std::vector<int64_t> prefix_displs(const std::vector<int64_t>& counts) {
   std::vector<int64_t> d(counts.size(), 0);
@@ -959,9 +968,6 @@ struct AlltoallBackward : public M4ANode {
   int64_t gaxis, saxis, numelem;
 };
 
-struct Interval {
-  int64_t src_off, dst_off, len;
-};
 } // namespace
 
 Tensor Communicator::Alltoall(const Tensor& input, int64_t gatheraxis,

@@ -68,3 +68,11 @@ def recv(tensor: torch.Tensor, source: int, tag: int = 0, comm=None):
 
 def join_dummies(loopthrough: torch.Tensor, dummies: List[torch.Tensor]):
     return _m.JoinDummies(loopthrough, dummies)
+
+
+def barrier(comm=None):
+    """Synchronize all ranks (not in the reference API; implemented as a
+    1-element allreduce, which is a barrier on both transports)."""
+    import torch as _t
+
+    _comm(comm).Allreduce(_t.zeros(1), _m.MPI_SUM)

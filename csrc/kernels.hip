@@ -44,6 +44,18 @@ struct SlabPack {
   SlabArgs s[N];
 };
 
+// The nontemporal builtins want native scalar/vector types; HIP's uint4 is
+// a class. Map each copy granule to an equivalent native type.
+typedef unsigned int v4u __attribute__((ext_vector_type(4)));
+template <typename T>
+struct NtVec {
+  using type = T;
+};
+template <>
+struct NtVec<uint4> {
+  using type = v4u;
+};
+
 // One chunk index c in [0, chunks_per_row): c < n_body copies granule c of
 // the phase-aligned body; c == n_body copies the head and tail bytes.
 template <typename VecT, int NSLABS>
@@ -64,9 +76,13 @@ __global__ __launch_bounds__(256) void slab_copy_kernel(SlabPack<NSLABS> pack) {
     if (head > a.row_b) head = a.row_b;
     const long long body = (a.row_b - head) / G;
     if (c < body) {
-      const VecT* sp = reinterpret_cast<const VecT*>(srow + head) + c;
-      VecT* dp = reinterpret_cast<VecT*>(drow + head) + c;
-      *dp = *sp;
+      using NT = typename NtVec<VecT>::type;
+      const NT* sp = reinterpret_cast<const NT*>(srow + head) + c;
+      NT* dp = reinterpret_cast<NT*>(drow + head) + c;
+      // nontemporal: each byte is touched exactly once (pure streaming),
+      // so bypass the L1/L2 allocation (cdna_hip_programming.md nt-weights:
+      // right for once-read streams)
+      __builtin_nontemporal_store(__builtin_nontemporal_load(sp), dp);
     } else {
       // head + tail bytes, done by the one extra chunk per row
       for (long long k = 0; k < head; ++k) drow[k] = srow[k];
@@ -173,13 +189,17 @@ __global__ __launch_bounds__(256) void slab_copy_shift_kernel(
     if (c >= vs && c < bodyv) {
       const char* sal = sbase - s + 16 * c;  // 16-aligned, in-bounds
       if (s == 0) {
-        *reinterpret_cast<uint4*>(drow + head + 16 * c) =
-            *reinterpret_cast<const uint4*>(sal);
+        __builtin_nontemporal_store(
+            __builtin_nontemporal_load(reinterpret_cast<const v4u*>(sal)),
+            reinterpret_cast<v4u*>(drow + head + 16 * c));
       } else {
+        // shared 16B lines between neighboring chunks: keep these cached
         const uint4 lo = *reinterpret_cast<const uint4*>(sal);
         const uint4 hi = *reinterpret_cast<const uint4*>(sal + 16);
-        *reinterpret_cast<uint4*>(drow + head + 16 * c) =
-            byte_window(lo, hi, s);
+        const uint4 w = byte_window(lo, hi, s);
+        __builtin_nontemporal_store(
+            *reinterpret_cast<const v4u*>(&w),
+            reinterpret_cast<v4u*>(drow + head + 16 * c));
       }
     } else if (c == bodyv) {
       // bytewise worker: head (+ the excluded chunk 0, if any) and tail
@@ -271,14 +291,20 @@ template <typename T, typename Op>
 __global__ __launch_bounds__(256) void bitwise_reduce_kernel(
     const T* __restrict__ in, T* __restrict__ out, long long chunk_elems,
     long long nranks) {
+  using NT = typename NtVec<T>::type;
   const long long stride = (long long)gridDim.x * blockDim.x;
   for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
        i < chunk_elems; i += stride) {
-    T acc = in[i];
+    // every element is touched once: nontemporal streaming
+    NT av = __builtin_nontemporal_load(reinterpret_cast<const NT*>(in) + i);
+    T acc = *reinterpret_cast<const T*>(&av);
     for (long long r = 1; r < nranks; ++r) {
-      acc = Op::apply(acc, in[r * chunk_elems + i]);
+      NT bv = __builtin_nontemporal_load(reinterpret_cast<const NT*>(in) +
+                                         r * chunk_elems + i);
+      acc = Op::apply(acc, *reinterpret_cast<const T*>(&bv));
     }
-    out[i] = acc;
+    __builtin_nontemporal_store(*reinterpret_cast<const NT*>(&acc),
+                                reinterpret_cast<NT*>(out) + i);
   }
 }
 
@@ -394,14 +420,16 @@ __global__ __launch_bounds__(256) void fp8_reduce_kernel(
   const long long stride = (long long)gridDim.x * blockDim.x;
   for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
        i < nvec; i += stride) {
-    const ulonglong1* base = reinterpret_cast<const ulonglong1*>(in);
-    unsigned long long w = base[i].x;
+    const unsigned long long* base =
+        reinterpret_cast<const unsigned long long*>(in);
+    unsigned long long w = __builtin_nontemporal_load(base + i);
     float acc[8];
     for (int k = 0; k < 8; ++k) {
       acc[k] = fp8_to_f32<E5M2>((unsigned char)(w >> (8 * k)));
     }
     for (long long r = 1; r < nranks; ++r) {
-      unsigned long long wr = base[r * nvec + i].x;  // n % 8 == 0 layout
+      unsigned long long wr =
+          __builtin_nontemporal_load(base + r * nvec + i);  // n % 8 == 0
       for (int k = 0; k < 8; ++k) {
         acc[k] = Op::apply(acc[k], fp8_to_f32<E5M2>((unsigned char)(wr >> (8 * k))));
       }
@@ -410,7 +438,8 @@ __global__ __launch_bounds__(256) void fp8_reduce_kernel(
     for (int k = 0; k < 8; ++k) {
       o |= (unsigned long long)f32_to_fp8<E5M2>(acc[k]) << (8 * k);
     }
-    reinterpret_cast<ulonglong1*>(out)[i].x = o;
+    __builtin_nontemporal_store(
+        o, reinterpret_cast<unsigned long long*>(out) + i);
   }
 }
 

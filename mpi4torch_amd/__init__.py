@@ -26,6 +26,8 @@ from typing import List, Optional
 
 import torch
 
+__version__ = "0.1.0"
+
 from . import _C  # registers torch.classes.mpi4torch_amd.* and torch.ops.mpi4torch_amd.*
 
 MPI_MAX: int = int(_C.MPI_MAX)

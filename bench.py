@@ -226,6 +226,15 @@ def main():
     assert comm.size == world
 
     result, t_step = BENCHES[args.config](m4a, comm, device, args)
+    # RCCL's version banner is written through C stdio, which is fully
+    # buffered on a pipe and flushes at exit — AFTER our JSON. Flush it now
+    # so the JSON line is the last thing on stdout.
+    try:
+        import ctypes
+
+        ctypes.CDLL(None).fflush(None)
+    except Exception:
+        pass
     if rank == 0:
         out = {
             "metric": result["metric"],

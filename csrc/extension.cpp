@@ -50,6 +50,7 @@ static auto communicator_class =
         .def("Allgather", &Communicator::Allgather)
         .def("Scatter", &Communicator::Scatter)
         .def("Alltoall", &Communicator::Alltoall)
+        .def("Alltoallv", &Communicator::Alltoallv)
         .def("Iallreduce", &Communicator::Iallreduce)
         .def("Isend", &Communicator::Isend)
         .def("Irecv", &Communicator::Irecv)

@@ -970,6 +970,25 @@ variable_list AllgatherBackward::apply(variable_list&& grads) {
 // ---------------------------------------------------------------------------
 
 namespace {
+struct AlltoallvBackward : public M4ANode {
+  AlltoallvBackward(int64_t gaxis, int64_t saxis,
+                    std::vector<int64_t> tgt, std::vector<int64_t> src)
+      : gaxis(gaxis), saxis(saxis), tgt(std::move(tgt)),
+        src(std::move(src)) {}
+  std::string name() const override { return "M4AAlltoallvBackward"; }
+  variable_list apply(variable_list&& grads) override {
+    variable_list out(1);
+    if (should_compute_output(0)) {
+      // adjoint: axes swapped, count vectors swapped (the gradient flows
+      // back to the original partition)
+      out[0] = comm->Alltoallv(grads[0], saxis, gaxis, tgt, src);
+    }
+    return out;
+  }
+  int64_t gaxis, saxis;
+  std::vector<int64_t> tgt, src;
+};
+
 struct AlltoallBackward : public M4ANode {
   AlltoallBackward(int64_t gaxis, int64_t saxis, int64_t numelem)
       : gaxis(gaxis), saxis(saxis), numelem(numelem) {}
@@ -988,15 +1007,24 @@ struct AlltoallBackward : public M4ANode {
 
 } // namespace
 
-Tensor Communicator::Alltoall(const Tensor& input, int64_t gatheraxis,
-                              int64_t scatteraxis, int64_t numelem) {
+Tensor Communicator::AlltoallvImpl(const Tensor& input, int64_t gatheraxis,
+                                   int64_t scatteraxis, int64_t numelem,
+                                   const std::vector<int64_t>& target_counts,
+                                   const std::vector<int64_t>& source_sizes) {
   gatheraxis = at::maybe_wrap_dim(gatheraxis, input.dim());
   scatteraxis = at::maybe_wrap_dim(scatteraxis, input.dim());
+  const bool explicit_counts = !target_counts.empty();
   std::shared_ptr<M4ANode> grad_fn;
   if (torch::autograd::compute_requires_grad(input)) {
-    grad_fn = make_node<AlltoallBackward>(
-        c10::intrusive_ptr<Communicator>::unsafe_reclaim_from_nonowning(this),
-        gatheraxis, scatteraxis, input.size(gatheraxis));
+    if (explicit_counts) {
+      grad_fn = make_node<AlltoallvBackward>(
+          c10::intrusive_ptr<Communicator>::unsafe_reclaim_from_nonowning(this),
+          gatheraxis, scatteraxis, source_sizes, target_counts);
+    } else {
+      grad_fn = make_node<AlltoallBackward>(
+          c10::intrusive_ptr<Communicator>::unsafe_reclaim_from_nonowning(this),
+          gatheraxis, scatteraxis, input.size(gatheraxis));
+    }
     grad_fn->set_next_edges(torch::autograd::collect_next_edges(input));
   }
   auto result = [&]() {
@@ -1005,7 +1033,8 @@ Tensor Communicator::Alltoall(const Tensor& input, int64_t gatheraxis,
     auto in = stager.to_comm(input).contiguous().variable_data();
     auto& tr = tr_for(in);
     if (w1_shortcut(tr)) {
-      TORCH_CHECK(numelem == in.size(scatteraxis),
+      const int64_t want = explicit_counts ? target_counts[0] : numelem;
+      TORCH_CHECK(want == in.size(scatteraxis),
                   "Alltoall: numelem must equal the scatter-axis size at "
                   "world size 1");
       return stager.from_comm(fast_clone(in));
@@ -1015,8 +1044,20 @@ Tensor Communicator::Alltoall(const Tensor& input, int64_t gatheraxis,
     debug_check_collective(group_name_, "Alltoall", at::empty({0}),
                            {gatheraxis, scatteraxis});
 
-    // What each rank wants along the scatter axis:
-    auto scounts = host_allgather_int64(group_name_, numelem);
+    // What each rank wants along the scatter axis. With explicit counts
+    // (Alltoallv) both vectors are caller-provided and the two host
+    // round-trips are skipped — the expert-parallel hot path, where every
+    // rank already knows the routing table.
+    std::vector<int64_t> scounts;
+    if (explicit_counts) {
+      TORCH_CHECK((int)target_counts.size() == P,
+                  "Alltoallv: target_counts must have world_size entries");
+      TORCH_CHECK((int)source_sizes.size() == P,
+                  "Alltoallv: source_sizes must have world_size entries");
+      scounts = target_counts;
+    } else {
+      scounts = host_allgather_int64(group_name_, numelem);
+    }
     auto sdispls = prefix_displs(scounts);
 
     std::vector<int64_t> s_offs(P), s_lens(P);  // my send slices (in `in`)
@@ -1029,7 +1070,13 @@ Tensor Communicator::Alltoall(const Tensor& input, int64_t gatheraxis,
       TORCH_CHECK(in.size(scatteraxis) == stotal,
                   "Alltoall: scatter-axis size (", in.size(scatteraxis),
                   ") must equal the sum of per-rank numelem (", stotal, ")");
-      auto gsizes = host_allgather_int64(group_name_, in.size(gatheraxis));
+      auto gsizes = explicit_counts
+                        ? source_sizes
+                        : host_allgather_int64(group_name_,
+                                               in.size(gatheraxis));
+      TORCH_CHECK(gsizes[me] == in.size(gatheraxis),
+                  "Alltoallv: source_sizes[rank] must equal the local "
+                  "gather-axis size");
       auto gdispls = prefix_displs(gsizes);
       for (int j = 0; j < P; ++j) {
         s_offs[j] = sdispls[j];
@@ -1042,7 +1089,13 @@ Tensor Communicator::Alltoall(const Tensor& input, int64_t gatheraxis,
     } else {
       // same-axis repartition (reference :947-979): ranks hold chunks of a
       // global axis; redistribute to the partition given by numelem
-      auto nsizes = host_allgather_int64(group_name_, in.size(gatheraxis));
+      auto nsizes = explicit_counts
+                        ? source_sizes
+                        : host_allgather_int64(group_name_,
+                                               in.size(gatheraxis));
+      TORCH_CHECK(nsizes[me] == in.size(gatheraxis),
+                  "Alltoallv: source_sizes[rank] must equal the local "
+                  "axis size");
       auto ndispls = prefix_displs(nsizes);
       const int64_t totaln = ndispls.back() + nsizes.back();
       const int64_t totalm = sdispls.back() + scounts.back();
@@ -1084,6 +1137,21 @@ Tensor Communicator::Alltoall(const Tensor& input, int64_t gatheraxis,
   }();
   attach_history(result, grad_fn);
   return result;
+}
+
+Tensor Communicator::Alltoall(const Tensor& input, int64_t gatheraxis,
+                              int64_t scatteraxis, int64_t numelem) {
+  return AlltoallvImpl(input, gatheraxis, scatteraxis, numelem, {}, {});
+}
+
+Tensor Communicator::Alltoallv(const Tensor& input, int64_t gatheraxis,
+                               int64_t scatteraxis,
+                               std::vector<int64_t> target_counts,
+                               std::vector<int64_t> source_sizes) {
+  TORCH_CHECK(!target_counts.empty(),
+              "Alltoallv: target_counts must not be empty");
+  return AlltoallvImpl(input, gatheraxis, scatteraxis,
+                       /*numelem=*/-1, target_counts, source_sizes);
 }
 
 // ---------------------------------------------------------------------------

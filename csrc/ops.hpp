@@ -36,6 +36,19 @@ struct Communicator : torch::CustomClassHolder {
                      int64_t numelem, int64_t root);
   at::Tensor Alltoall(const at::Tensor& input, int64_t gatheraxis,
                       int64_t scatteraxis, int64_t numelem);
+  // Explicit-counts variant (MI355X extension; not in the reference):
+  // skips the two host count exchanges when the caller already knows every
+  // rank's counts (expert-parallel routing). target_counts[j] = what rank j
+  // keeps along scatteraxis; source_sizes[j] = rank j's current
+  // gather/partition-axis size.
+  at::Tensor Alltoallv(const at::Tensor& input, int64_t gatheraxis,
+                       int64_t scatteraxis,
+                       std::vector<int64_t> target_counts,
+                       std::vector<int64_t> source_sizes);
+  at::Tensor AlltoallvImpl(const at::Tensor& input, int64_t gatheraxis,
+                           int64_t scatteraxis, int64_t numelem,
+                           const std::vector<int64_t>& target_counts,
+                           const std::vector<int64_t>& source_sizes);
 
   // Non-blocking allreduce (MI355X-first overlap primitive, not in the
   // reference API): returns a wait handle; no autograd through it — use

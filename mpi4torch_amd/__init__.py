@@ -184,6 +184,18 @@ class MPI_Communicator:
         """
         return self._comm.Alltoall(tensor, gatheraxis, scatteraxis, numelem)
 
+    def Alltoallv(self, tensor: torch.Tensor, gatheraxis: int,
+                  scatteraxis: int, target_counts: List[int],
+                  source_sizes: List[int]) -> torch.Tensor:
+        """Alltoall with caller-provided counts (MI355X extension): skips
+        the host count exchanges when every rank already knows the routing
+        (expert parallelism). ``target_counts[j]`` = slices rank j keeps
+        along ``scatteraxis``; ``source_sizes[j]`` = rank j's current
+        gather/partition-axis size. Backward: Alltoallv with axes and
+        count vectors swapped."""
+        return self._comm.Alltoallv(tensor, gatheraxis, scatteraxis,
+                                    target_counts, source_sizes)
+
     def Iallreduce(self, tensor: torch.Tensor, op: int) -> WaitHandle:
         """Non-blocking Allreduce (no autograd): returns a WaitHandle whose
         Wait() yields the reduced tensor. The overlap primitive behind

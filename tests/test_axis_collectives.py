@@ -210,3 +210,39 @@ def test_alltoall_ws5():
 
 def test_alltoall_ws7():
     run_spmd(7, _alltoall_worker)
+
+
+def _alltoallv_worker(rank, world):
+    import mpi4torch_amd as m
+
+    comm = m.COMM_WORLD
+    # explicit counts must agree exactly with what the implicit op computes
+    tot = world * (world + 1) // 2
+    tmp = torch.rand([3, rank + 1, 4, tot, 2], dtype=torch.double).requires_grad_()
+    tmp2 = tmp.detach().clone().requires_grad_()
+
+    target = [r + 1 for r in range(world)]
+    source = [r + 1 for r in range(world)]
+    r1 = comm.Alltoall(tmp, 1, 3, rank + 1)
+    r2 = comm.Alltoallv(tmp2, 1, 3, target, source)
+    assert torch.equal(r1, r2)
+    r1.sum().backward()
+    r2.sum().backward()
+    assert torch.equal(tmp.grad, tmp2.grad)
+
+    # same-axis repartition with explicit counts
+    x = torch.rand([2, rank + 2, 3], dtype=torch.double).requires_grad_()
+    src = [r + 2 for r in range(world)]
+    tgt = list(reversed(src))
+    out = comm.Alltoallv(x, 1, 1, tgt, src)
+    assert out.shape[1] == tgt[rank]
+    out.sum().backward()
+    assert (x.grad == torch.ones_like(x)).all()
+
+
+def test_alltoallv_ws2():
+    run_spmd(2, _alltoallv_worker)
+
+
+def test_alltoallv_ws5():
+    run_spmd(5, _alltoallv_worker)

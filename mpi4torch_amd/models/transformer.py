@@ -47,8 +47,11 @@ class UlyssesTransformerBlock(torch.nn.Module):
         qkv = self.qkv(h).view(B, S_local, 3, self.n_heads, self.head_dim)
         if self.comm.size > 1:
             # [B, S/P, 3, H, d] -> [B, S, 3, H/P, d]: full sequence,
-            # sharded heads (one Alltoall for q, k and v together)
-            qkv = self.comm.Alltoall(qkv, 1, 3, self.n_heads // self.comm.size)
+            # sharded heads (one Alltoallv for q, k and v together; counts
+            # are uniform and locally known — no host exchange)
+            P = self.comm.size
+            qkv = self.comm.Alltoallv(qkv, 1, 3, [self.n_heads // P] * P,
+                                      [S_local] * P)
         q, k, v = qkv.unbind(2)
         attn = torch.nn.functional.scaled_dot_product_attention(
             q.transpose(1, 2), k.transpose(1, 2), v.transpose(1, 2)

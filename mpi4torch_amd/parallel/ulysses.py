@@ -25,12 +25,17 @@ def ulysses_reshard(t: torch.Tensor, gather_axis: int, scatter_axis: int,
 def seq_to_head(t: torch.Tensor, comm=None, seq_axis: int = 1,
                 head_axis: int = 2) -> torch.Tensor:
     """[b, s/P, h, d] -> [b, s, h/P, d]: full sequence, sharded heads —
-    the layout attention wants."""
+    the layout attention wants. Uses the explicit-counts Alltoallv (all
+    counts are uniform and locally known), skipping the host exchanges."""
     comm = comm if comm is not None else m4a.COMM_WORLD
     heads = t.size(head_axis)
     assert heads % comm.size == 0, (
         f"head count {heads} must divide by world size {comm.size}")
-    return comm.Alltoall(t, seq_axis, head_axis, heads // comm.size)
+    if comm.size == 1:
+        return comm.Alltoall(t, seq_axis, head_axis, heads)
+    target = [heads // comm.size] * comm.size
+    source = [t.size(seq_axis)] * comm.size
+    return comm.Alltoallv(t, seq_axis, head_axis, target, source)
 
 
 def head_to_seq(t: torch.Tensor, comm=None, seq_axis: int = 1,
@@ -40,4 +45,8 @@ def head_to_seq(t: torch.Tensor, comm=None, seq_axis: int = 1,
     seq = t.size(seq_axis)
     assert seq % comm.size == 0, (
         f"sequence length {seq} must divide by world size {comm.size}")
-    return comm.Alltoall(t, head_axis, seq_axis, seq // comm.size)
+    if comm.size == 1:
+        return comm.Alltoall(t, head_axis, seq_axis, seq)
+    target = [seq // comm.size] * comm.size
+    source = [t.size(head_axis)] * comm.size
+    return comm.Alltoallv(t, head_axis, seq_axis, target, source)

@@ -51,6 +51,7 @@ static auto communicator_class =
         .def("Scatter", &Communicator::Scatter)
         .def("Alltoall", &Communicator::Alltoall)
         .def("Alltoallv", &Communicator::Alltoallv)
+        .def("AlltoallPairwise", &Communicator::AlltoallPairwise)
         .def("Iallreduce", &Communicator::Iallreduce)
         .def("Isend", &Communicator::Isend)
         .def("Irecv", &Communicator::Irecv)

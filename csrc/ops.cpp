@@ -1154,6 +1154,101 @@ Tensor Communicator::Alltoallv(const Tensor& input, int64_t gatheraxis,
                        /*numelem=*/-1, target_counts, source_sizes);
 }
 
+
+// ---------------------------------------------------------------------------
+// AlltoallPairwise — the expert-parallel dispatch/combine primitive
+// (MI355X extension; the reference's same-axis Alltoall can only
+// repartition contiguous global intervals, reference :947-979).
+// ---------------------------------------------------------------------------
+
+namespace {
+struct AlltoallPairwiseBackward : public M4ANode {
+  AlltoallPairwiseBackward(int64_t axis, std::vector<int64_t> send,
+                           std::vector<int64_t> recv)
+      : axis(axis), send(std::move(send)), recv(std::move(recv)) {}
+  std::string name() const override { return "M4AAlltoallPairwiseBackward"; }
+  variable_list apply(variable_list&& grads) override {
+    variable_list out(1);
+    if (should_compute_output(0)) {
+      // adjoint: return every received slice to its sender — the count
+      // matrix transposes, i.e. swap send/recv vectors
+      out[0] = comm->AlltoallPairwise(grads[0], axis, recv, send);
+    }
+    return out;
+  }
+  int64_t axis;
+  std::vector<int64_t> send, recv;
+};
+} // namespace
+
+Tensor Communicator::AlltoallPairwise(const Tensor& input, int64_t axis,
+                                      std::vector<int64_t> send_counts,
+                                      std::vector<int64_t> recv_counts) {
+  axis = at::maybe_wrap_dim(axis, input.dim());
+  auto& tr0 = cpu_tr();
+  const int P = tr0.size();
+  TORCH_CHECK((int)send_counts.size() == P,
+              "AlltoallPairwise: send_counts must have world_size entries");
+  if (recv_counts.empty() && P > 1) {
+    // exchange the count matrix; my recv from j = j's send to me
+    auto mat = host_allgather_int64_vec(group_name_, send_counts);
+    recv_counts.resize(P);
+    const int me = tr0.rank();
+    for (int j = 0; j < P; ++j) recv_counts[j] = mat[(int64_t)j * P + me];
+  }
+  if (P == 1 && recv_counts.empty()) recv_counts = send_counts;
+  TORCH_CHECK((int)recv_counts.size() == P,
+              "AlltoallPairwise: recv_counts must have world_size entries");
+
+  std::shared_ptr<M4ANode> grad_fn;
+  if (torch::autograd::compute_requires_grad(input)) {
+    grad_fn = make_node<AlltoallPairwiseBackward>(
+        c10::intrusive_ptr<Communicator>::unsafe_reclaim_from_nonowning(this),
+        axis, send_counts, recv_counts);
+    grad_fn->set_next_edges(torch::autograd::collect_next_edges(input));
+  }
+  auto result = [&]() {
+    at::AutoDispatchBelowADInplaceOrView guard;
+    DeviceStager stager(input);
+    auto in = stager.to_comm(input).contiguous().variable_data();
+    auto& tr = tr_for(in);
+    auto s_displs = prefix_displs(send_counts);
+    auto r_displs = prefix_displs(recv_counts);
+    const int64_t s_total = s_displs.back() + send_counts.back();
+    const int64_t r_total = r_displs.back() + recv_counts.back();
+    TORCH_CHECK(in.size(axis) == s_total,
+                "AlltoallPairwise: sum of send_counts (", s_total,
+                ") must equal the axis size (", in.size(axis), ")");
+    auto outsizes = in.sizes().vec();
+    outsizes[axis] = r_total;
+    auto out = at::empty(outsizes, in.options());
+    if (tr.size() == 1 && !config().force_full_path) {
+      // world of one: send_counts == recv_counts, identity
+      out.copy_(in);
+      return stager.from_comm(std::move(out));
+    }
+    const auto g = axis_geom(in, axis);
+    auto sblocks = make_blocks(in, g.before, g.after, send_counts, s_displs,
+                               &in);
+    if (g.before != 1) {
+      move_axis_blocks(in, axis, s_displs, send_counts, sblocks,
+                       /*pack=*/true);
+    }
+    const auto gr = axis_geom(out, axis);
+    auto rblocks = make_blocks(out, gr.before, gr.after, recv_counts,
+                               r_displs, &out);
+    auto peers = iota_peers(tr.size());
+    tr.exchange(sblocks, peers, rblocks, peers);
+    if (gr.before != 1) {
+      move_axis_blocks(out, axis, r_displs, recv_counts, rblocks,
+                       /*pack=*/false);
+    }
+    return stager.from_comm(std::move(out));
+  }();
+  attach_history(result, grad_fn);
+  return result;
+}
+
 // ---------------------------------------------------------------------------
 // Isend / Irecv / Wait (reference csrc/extension.cpp:1048-1265). The handle
 // is the same 3-tensor contract [meta, buffer, input]; the MPI_Request

@@ -49,6 +49,15 @@ struct Communicator : torch::CustomClassHolder {
                            int64_t scatteraxis, int64_t numelem,
                            const std::vector<int64_t>& target_counts,
                            const std::vector<int64_t>& source_sizes);
+  // Pairwise-count alltoall (MI355X extension): rank r sends
+  // send_counts[j] slices of `axis` to rank j (arbitrary P x P count
+  // matrix — the expert-parallel token dispatch, inexpressible as the
+  // reference's interval repartition). recv_counts, if empty, are
+  // exchanged over the host channel. Backward sends every received slice
+  // back (the exact adjoint).
+  at::Tensor AlltoallPairwise(const at::Tensor& input, int64_t axis,
+                              std::vector<int64_t> send_counts,
+                              std::vector<int64_t> recv_counts);
 
   // Non-blocking allreduce (MI355X-first overlap primitive, not in the
   // reference API): returns a wait handle; no autograd through it — use

@@ -838,6 +838,25 @@ std::vector<int64_t> host_allgather_int64(const std::string& group_name,
   return res;
 }
 
+std::vector<int64_t> host_allgather_int64_vec(
+    const std::string& group_name, const std::vector<int64_t>& v) {
+  auto backend = gloo_backend(group_name);
+  const int n = backend->getSize();
+  const int64_t len = (int64_t)v.size();
+  auto in = at::empty({len}, at::TensorOptions().dtype(at::kLong));
+  std::memcpy(in.data_ptr(), v.data(), len * sizeof(int64_t));
+  auto out = at::empty({n * len}, at::TensorOptions().dtype(at::kLong));
+  std::vector<at::Tensor> outs;
+  outs.reserve(n);
+  for (int i = 0; i < n; ++i) outs.push_back(out.narrow(0, i * len, len));
+  std::vector<std::vector<at::Tensor>> outputs{outs};
+  std::vector<at::Tensor> inputs{in};
+  backend->allgather(outputs, inputs)->wait();
+  std::vector<int64_t> res(n * len);
+  std::memcpy(res.data(), out.data_ptr(), n * len * sizeof(int64_t));
+  return res;
+}
+
 std::vector<int64_t> host_broadcast_int64(const std::string& group_name,
                                           const std::vector<int64_t>& values,
                                           int root, int64_t fixed_len) {

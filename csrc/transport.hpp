@@ -92,6 +92,11 @@ std::shared_ptr<Transport> make_rccl_transport(const std::string& group_name,
 // replaces reference's MPI_Gather/MPI_Bcast of ints for count exchanges).
 std::vector<int64_t> host_allgather_int64(const std::string& group_name,
                                           int64_t value);
+// Allgather a fixed-length int64 vector: returns rank-major concatenation
+// (P * len entries). Carries the per-pair count matrices of the pairwise
+// alltoall.
+std::vector<int64_t> host_allgather_int64_vec(const std::string& group_name,
+                                              const std::vector<int64_t>& v);
 std::vector<int64_t> host_broadcast_int64(const std::string& group_name,
                                           const std::vector<int64_t>& values,
                                           int root, int64_t fixed_len);

@@ -196,6 +196,19 @@ class MPI_Communicator:
         return self._comm.Alltoallv(tensor, gatheraxis, scatteraxis,
                                     target_counts, source_sizes)
 
+    def AlltoallPairwise(self, tensor: torch.Tensor, axis: int,
+                         send_counts: List[int],
+                         recv_counts: List[int]) -> torch.Tensor:
+        """Arbitrary pairwise-count alltoall along ``axis`` (MI355X
+        extension): this rank sends ``send_counts[j]`` slices to rank j.
+        The expert-parallel token dispatch/combine primitive — the
+        reference's same-axis Alltoall only repartitions contiguous global
+        intervals. Pass ``recv_counts=[]`` to have them exchanged
+        automatically. Backward returns every received slice to its
+        sender (exact adjoint)."""
+        return self._comm.AlltoallPairwise(tensor, axis, send_counts,
+                                           recv_counts)
+
     def Iallreduce(self, tensor: torch.Tensor, op: int) -> WaitHandle:
         """Non-blocking Allreduce (no autograd): returns a WaitHandle whose
         Wait() yields the reduced tensor. The overlap primitive behind

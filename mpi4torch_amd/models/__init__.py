@@ -7,5 +7,6 @@ doc/examples.rst). These modules are those patterns as reusable code.
 
 from .linreg import DistributedLinReg
 from .transformer import UlyssesTransformerBlock
+from .moe import ExpertParallelMoE
 
-__all__ = ["DistributedLinReg", "UlyssesTransformerBlock"]
+__all__ = ["DistributedLinReg", "UlyssesTransformerBlock", "ExpertParallelMoE"]

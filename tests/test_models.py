@@ -54,3 +54,56 @@ def test_ulysses_block_ws2():
 
 def test_ulysses_block_ws4():
     run_spmd(4, _ulysses_block_worker)
+
+
+def _moe_worker(rank, world):
+    import mpi4torch_amd as m
+    from mpi4torch_amd.models.moe import ExpertParallelMoE
+
+    torch.manual_seed(3)  # identical router + expert weights on all ranks
+    d_model, n_experts, N = 8, 2 * world, 12
+    moe = ExpertParallelMoE(d_model, n_experts, d_hidden=16).double()
+
+    torch.manual_seed(50 + rank)
+    x = torch.randn(N, d_model, dtype=torch.double).requires_grad_()
+
+    y = moe(x)
+    y.square().sum().backward()
+
+    # dense reference computed locally: same router, and the experts of
+    # EVERY rank reconstructed with the same seed stream as each rank's
+    # module init (seed 3 creates router then world*eppr experts in order,
+    # identical on every rank because n_experts/d are identical)
+    torch.manual_seed(3)
+    full = ExpertParallelMoE(d_model, n_experts, d_hidden=16).double()
+
+    x_ref = x.detach().clone().requires_grad_()
+    logits = full.router(x_ref)
+    gates = torch.softmax(logits, dim=-1)
+    expert = torch.argmax(gates, dim=-1)
+    gate = gates.gather(1, expert.unsqueeze(1)).squeeze(1)
+    y_ref = torch.zeros_like(x_ref)
+    # full.experts holds only this world's shard; rebuild every rank's
+    # experts from their shared init: all ranks created the SAME
+    # experts_per_rank modules (same seed), i.e. expert e on rank r equals
+    # local module (e % experts_per_rank)! Route accordingly.
+    for gid in range(n_experts):
+        mask = expert == gid
+        if bool(mask.any()):
+            idx = mask.nonzero(as_tuple=True)[0]
+            mod = full.experts[gid % full.experts_per_rank]
+            y_ref = y_ref.index_copy(0, idx, mod(x_ref.index_select(0, idx)))
+    y_ref = y_ref * gate.unsqueeze(1)
+    ref_loss = y_ref.square().sum()
+    ref_loss.backward()
+
+    assert torch.allclose(y, y_ref, atol=1e-9), (y - y_ref).abs().max()
+    assert torch.allclose(x.grad, x_ref.grad, atol=1e-9)
+
+
+def test_moe_ws2():
+    run_spmd(2, _moe_worker)
+
+
+def test_moe_ws4():
+    run_spmd(4, _moe_worker)

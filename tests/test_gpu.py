@@ -283,3 +283,20 @@ def test_full_path_axis_collectives(world1):
         torch.testing.assert_close(y.grad, torch.ones_like(y))
     finally:
         m._C.force_full_path(False)
+
+
+def test_alltoall_pairwise_gpu(world1):
+    m = world1
+    comm = m.COMM_WORLD
+    x = torch.rand(2, 9, 3, device="cuda", dtype=torch.bfloat16)
+    out = comm.AlltoallPairwise(x, 1, [9], [])
+    torch.testing.assert_close(out.float(), x.float())
+    m._C.force_full_path(True)
+    try:
+        y = torch.rand(4, 12, device="cuda").requires_grad_()
+        out2 = comm.AlltoallPairwise(y, 0, [4], [4])
+        torch.testing.assert_close(out2, y.detach())
+        out2.sum().backward()
+        torch.testing.assert_close(y.grad, torch.ones_like(y))
+    finally:
+        m._C.force_full_path(False)

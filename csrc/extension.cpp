@@ -48,6 +48,7 @@ static auto communicator_class =
         .def("Reduce_", &Communicator::Reduce_)
         .def("Gather", &Communicator::Gather)
         .def("Allgather", &Communicator::Allgather)
+        .def("Reducescatter", &Communicator::Reducescatter)
         .def("Scatter", &Communicator::Scatter)
         .def("Alltoall", &Communicator::Alltoall)
         .def("Alltoallv", &Communicator::Alltoallv)

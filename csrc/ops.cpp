@@ -963,6 +963,90 @@ variable_list AllgatherBackward::apply(variable_list&& grads) {
 } // namespace
 
 // ---------------------------------------------------------------------------
+// Reducescatter — Allgather's adjoint as a public op (MI355X extension).
+// ---------------------------------------------------------------------------
+
+namespace {
+struct ReducescatterBackward : public M4ANode {
+  explicit ReducescatterBackward(int64_t axis) : axis(axis) {}
+  std::string name() const override { return "M4AReducescatterBackward"; }
+  variable_list apply(variable_list&& grads) override {
+    variable_list out(1);
+    if (should_compute_output(0)) {
+      // every rank's input slice region sees exactly the gradient of the
+      // rank that kept it: concat = allgather
+      out[0] = comm->Allgather(grads[0], axis);
+    }
+    return out;
+  }
+  int64_t axis;
+};
+} // namespace
+
+Tensor Communicator::Reducescatter(const Tensor& input, int64_t axis,
+                                   int64_t numelem) {
+  axis = at::maybe_wrap_dim(axis, input.dim());
+  std::shared_ptr<M4ANode> grad_fn;
+  if (torch::autograd::compute_requires_grad(input)) {
+    grad_fn = make_node<ReducescatterBackward>(
+        c10::intrusive_ptr<Communicator>::unsafe_reclaim_from_nonowning(this),
+        axis);
+    grad_fn->set_next_edges(torch::autograd::collect_next_edges(input));
+  }
+  auto result = [&]() {
+    at::AutoDispatchBelowADInplaceOrView guard;
+    DeviceStager stager(input);
+    auto in = stager.to_comm(input).contiguous().variable_data();
+    auto& tr = tr_for(in);
+    if (w1_shortcut(tr)) {
+      TORCH_CHECK(numelem == in.size(axis),
+                  "Reducescatter: numelem must equal the axis size at "
+                  "world size 1");
+      return stager.from_comm(fast_clone(in));
+    }
+    debug_check_collective(group_name_, "Reducescatter", at::empty({0}),
+                           {axis, numelem});
+    const auto g = axis_geom(in, axis);
+    auto counts = host_allgather_int64(group_name_, numelem);
+    auto displs = prefix_displs(counts);
+    const int64_t total = displs.back() + counts.back();
+    TORCH_CHECK(total == g.axis,
+                "Reducescatter: sum of per-rank numelem (", total,
+                ") must equal the axis size (", g.axis, ")");
+    const int me = tr.rank();
+    auto outsizes = in.sizes().vec();
+    outsizes[axis] = counts[me];
+    const bool equal = std::all_of(counts.begin(), counts.end(),
+                                   [&](int64_t c) { return c == counts[0]; });
+    if (equal && tr.is_gpu() && native_reduce_dtype(tr, in.scalar_type())) {
+      auto res = at::empty(outsizes, in.options());
+      auto res_flat = res.view({-1});
+      if (g.before == 1) {
+        tr.reduce_scatter_equal(in, res_flat, kSum);
+      } else {
+        const int64_t chunk = g.before * counts[0] * g.after;
+        auto staging = at::empty({(int64_t)counts.size() * chunk},
+                                 in.options());
+        std::vector<Tensor> blocks(counts.size());
+        for (size_t r = 0; r < counts.size(); ++r) {
+          blocks[r] = staging.narrow(0, (int64_t)r * chunk, chunk);
+        }
+        move_axis_blocks(in, axis, displs, counts, blocks, /*pack=*/true);
+        tr.reduce_scatter_equal(staging, res_flat, kSum);
+      }
+      return stager.from_comm(std::move(res));
+    }
+    // general path: allreduce then slice (CPU, variable counts, exotic
+    // dtypes incl. fp8 via the lowered allreduce)
+    auto red = allreduce_lowered(tr, in, kSum);
+    return stager.from_comm(
+        red.narrow(axis, displs[me], counts[me]).contiguous());
+  }();
+  attach_history(result, grad_fn);
+  return result;
+}
+
+// ---------------------------------------------------------------------------
 // Alltoall (reference csrc/extension.cpp:886-987). Unlike the reference's
 // composite of GetSize() successive Scatters (ref :940-981, which serializes
 // P collectives), this is ONE grouped RCCL p2p exchange with fused CDNA4

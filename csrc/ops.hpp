@@ -32,6 +32,13 @@ struct Communicator : torch::CustomClassHolder {
   at::Tensor Reduce_(const at::Tensor& input, int64_t op, int64_t root);
   at::Tensor Gather(const at::Tensor& input, int64_t gatheraxis, int64_t root);
   at::Tensor Allgather(const at::Tensor& input, int64_t gatheraxis);
+  // Adjoint pair of Allgather, exposed as a public collective (MI355X
+  // extension; not in the reference API): elementwise-SUM across ranks,
+  // then this rank keeps `numelem` slices of `axis` (per-rank counts may
+  // differ). Backward: Allgather of the gradient. Enables ZeRO-style
+  // gradient/optimizer sharding (parallel/zero.py).
+  at::Tensor Reducescatter(const at::Tensor& input, int64_t axis,
+                           int64_t numelem);
   at::Tensor Scatter(const at::Tensor& input, int64_t scatteraxis,
                      int64_t numelem, int64_t root);
   at::Tensor Alltoall(const at::Tensor& input, int64_t gatheraxis,

@@ -167,6 +167,14 @@ class MPI_Communicator:
         """
         return self._comm.Allgather(tensor, gatheraxis)
 
+    def Reducescatter(self, tensor: torch.Tensor, axis: int,
+                      numelem: int) -> torch.Tensor:
+        """Elementwise-SUM across ranks, keeping ``numelem`` slices of
+        ``axis`` locally (MI355X extension; the exact adjoint pair of
+        Allgather — its backward IS Allgather). The ZeRO gradient-sharding
+        primitive."""
+        return self._comm.Reducescatter(tensor, axis, numelem)
+
     def Scatter(self, tensor: torch.Tensor, scatteraxis: int, numelem: int,
                 root: int) -> torch.Tensor:
         """Distribute ``root``'s tensor along ``scatteraxis``; this rank

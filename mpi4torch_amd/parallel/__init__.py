@@ -11,6 +11,7 @@ from .ulysses import ulysses_reshard, seq_to_head, head_to_seq
 from .tp import (ColumnParallelLinear, RowParallelLinear, TensorParallelMLP,
                  copy_to_tp_region)
 from .pipeline import GPipe
+from .zero import ZeroRedundancyOptimizer
 
 __all__ = [
     "DistributedDataParallel",
@@ -22,4 +23,5 @@ __all__ = [
     "TensorParallelMLP",
     "copy_to_tp_region",
     "GPipe",
+    "ZeroRedundancyOptimizer",
 ]

@@ -1,0 +1,93 @@
+"""ZeRO-1: optimizer-state sharding over Reducescatter / Allgather.
+
+Each data-parallel rank owns 1/P of the flattened parameter vector and
+runs the inner optimizer only on that shard — optimizer state (Adam
+moments etc.) shrinks by P. Per step:
+
+1. gradients are flattened and Reducescatter'ed (sum + shard in one
+   collective — half the wire bytes of allreduce-then-slice),
+2. the inner optimizer updates the local shard,
+3. the updated shard is Allgather'ed back into the full parameters.
+
+Sized for MI355X: one large flat collective per step (288 GB HBM3E makes
+the flat buffer free; xGMI rings want few, large transfers).
+"""
+
+from typing import Iterable, List
+
+import torch
+
+import mpi4torch_amd as m4a
+
+
+class ZeroRedundancyOptimizer:
+    def __init__(self, params: Iterable[torch.nn.Parameter], optimizer_cls,
+                 comm=None, average: bool = True, **optim_kwargs):
+        self.comm = comm if comm is not None else m4a.COMM_WORLD
+        self.params: List[torch.nn.Parameter] = [
+            p for p in params if p.requires_grad
+        ]
+        assert self.params, "no trainable parameters"
+        self.average = average
+        P = self.comm.size
+        self._numels = [p.numel() for p in self.params]
+        total = sum(self._numels)
+        self._shard_len = (total + P - 1) // P  # equal shards, padded
+        self._padded = self._shard_len * P
+
+        dev = self.params[0].device
+        dt = self.params[0].dtype
+        with torch.no_grad():
+            flat = torch.zeros(self._padded, device=dev, dtype=dt)
+            torch.cat([p.reshape(-1) for p in self.params],
+                      out=flat[:total])
+        lo = self.comm.rank * self._shard_len
+        self._shard = flat[lo:lo + self._shard_len].clone().requires_grad_()
+        self.optimizer = optimizer_cls([self._shard], **optim_kwargs)
+
+    def zero_grad(self, set_to_none: bool = True):
+        for p in self.params:
+            if set_to_none:
+                p.grad = None
+            elif p.grad is not None:
+                p.grad.zero_()
+
+    @torch.no_grad()
+    def step(self):
+        P = self.comm.size
+        total = sum(self._numels)
+        dev = self._shard.device
+        gflat = torch.zeros(self._padded, device=dev,
+                            dtype=self._shard.dtype)
+        off = 0
+        for p, n in zip(self.params, self._numels):
+            if p.grad is not None:
+                gflat[off:off + n].copy_(p.grad.reshape(-1))
+            off += n
+        # sum + shard in ONE collective (the adjoint pair of the final
+        # allgather); average for the usual DP convention
+        gshard = self.comm.Reducescatter(gflat, 0, self._shard_len)
+        if self.average and P > 1:
+            gshard = gshard / P
+        self._shard.grad = gshard
+        self.optimizer.step()
+        self._shard.grad = None
+        # materialize updated parameters everywhere
+        full = (self.comm.Allgather(self._shard.detach(), 0)
+                if P > 1 else self._shard.detach())
+        off = 0
+        for p, n in zip(self.params, self._numels):
+            p.data.copy_(full[off:off + n].view_as(p))
+            off += n
+
+    def state_dict(self):
+        return {
+            "inner": self.optimizer.state_dict(),
+            "shard": self._shard.detach().cpu(),
+            "rank": self.comm.rank,
+        }
+
+    def load_state_dict(self, state):
+        self.optimizer.load_state_dict(state["inner"])
+        with torch.no_grad():
+            self._shard.copy_(state["shard"].to(self._shard.device))

@@ -300,3 +300,42 @@ def test_alltoall_pairwise_ws2():
 
 def test_alltoall_pairwise_ws5():
     run_spmd(5, _pairwise_worker)
+
+
+def _reducescatter_worker(rank, world):
+    import mpi4torch_amd as m
+
+    comm = m.COMM_WORLD
+    # forward: sum across ranks, slice per rank (variable counts)
+    counts = [r + 1 for r in range(world)]
+    total = sum(counts)
+    x = torch.zeros(2, total, 3, dtype=torch.double)
+    x[:] = rank + 1
+    out = comm.Reducescatter(x, 1, counts[rank])
+    assert list(out.shape) == [2, counts[rank], 3]
+    assert (out == world * (world + 1) / 2).all()
+
+    # adjoint: backward of reduce-scatter is allgather — a NON-uniform
+    # gradient must land replicated on every rank's full input
+    x2 = torch.rand(2, total, 3, dtype=torch.double).requires_grad_()
+    out2 = comm.Reducescatter(x2, 1, counts[rank])
+    (out2 * (rank + 1)).sum().backward()
+    expect = torch.empty_like(x2)
+    off = 0
+    for r in range(world):
+        expect[:, off:off + counts[r], :] = r + 1
+        off += counts[r]
+    assert torch.equal(x2.grad, expect)
+
+    # identity law: Reducescatter(Allgather(t)) == world * t
+    t = torch.rand(3, rank + 2, 2, dtype=torch.double)
+    rs = comm.Reducescatter(comm.Allgather(t, 1), 1, rank + 2)
+    assert torch.allclose(rs, world * t)
+
+
+def test_reducescatter_ws2():
+    run_spmd(2, _reducescatter_worker)
+
+
+def test_reducescatter_ws5():
+    run_spmd(5, _reducescatter_worker)

@@ -300,3 +300,20 @@ def test_alltoall_pairwise_gpu(world1):
         torch.testing.assert_close(y.grad, torch.ones_like(y))
     finally:
         m._C.force_full_path(False)
+
+
+def test_reducescatter_gpu(world1):
+    m = world1
+    comm = m.COMM_WORLD
+    x = torch.rand(4, 16, device="cuda", dtype=torch.bfloat16)
+    out = comm.Reducescatter(x, 1, 16)
+    torch.testing.assert_close(out.float(), x.float())
+    m._C.force_full_path(True)
+    try:
+        y = torch.rand(2, 8, 3, device="cuda").requires_grad_()
+        out2 = comm.Reducescatter(y, 1, 8)  # real ncclReduceScatter, P=1
+        torch.testing.assert_close(out2, y.detach())
+        out2.sum().backward()
+        torch.testing.assert_close(y.grad, torch.ones_like(y))
+    finally:
+        m._C.force_full_path(False)

@@ -46,6 +46,28 @@ def alltoall(tensor: torch.Tensor, gatheraxis: int, scatteraxis: int,
     return _comm(comm).Alltoall(tensor, gatheraxis, scatteraxis, numelem)
 
 
+def reducescatter(tensor: torch.Tensor, axis: int = 0, numelem: int = 1,
+                  comm=None):
+    return _comm(comm).Reducescatter(tensor, axis, numelem)
+
+
+def alltoallv(tensor: torch.Tensor, gatheraxis: int, scatteraxis: int,
+              target_counts, source_sizes, comm=None):
+    return _comm(comm).Alltoallv(tensor, gatheraxis, scatteraxis,
+                                 list(target_counts), list(source_sizes))
+
+
+def alltoall_pairwise(tensor: torch.Tensor, axis: int, send_counts,
+                      recv_counts=None, comm=None):
+    return _comm(comm).AlltoallPairwise(
+        tensor, axis, list(send_counts),
+        [] if recv_counts is None else list(recv_counts))
+
+
+def iallreduce(tensor: torch.Tensor, op: Optional[int] = None, comm=None):
+    return _comm(comm).Iallreduce(tensor, _m.MPI_SUM if op is None else op)
+
+
 def isend(tensor: torch.Tensor, dest: int, tag: int = 0, comm=None):
     return _comm(comm).Isend(tensor, dest, tag)
 

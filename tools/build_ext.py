@@ -65,6 +65,7 @@ def build(force: bool = False, verbose: bool = True) -> str:
         os.path.getmtime(os.path.join(CSRC, h)) for h in HEADERS
     )
     self_mtime = os.path.getmtime(os.path.abspath(__file__))
+    jobs = []
     for src in SOURCES:
         spath = os.path.join(CSRC, src)
         obj = os.path.join(BUILD, src.replace("/", "_") + ".o")
@@ -83,7 +84,10 @@ def build(force: bool = False, verbose: bool = True) -> str:
         cmd.append(f"--offload-arch={GPU_ARCH}")
         if verbose:
             print("[build_ext]", " ".join(cmd), flush=True)
-        subprocess.check_call(cmd, cwd=REPO)
+        jobs.append(subprocess.Popen(cmd, cwd=REPO))
+    failed = [j.wait() for j in jobs]  # TUs are independent: compile parallel
+    if any(failed):
+        raise RuntimeError("hipcc compilation failed")
 
     if (
         force

@@ -28,9 +28,9 @@ net = torch.nn.Sequential(
 zopt = ZeroRedundancyOptimizer(net.parameters(), torch.optim.AdamW, lr=1e-3)
 
 torch.manual_seed(900 + comm.rank)  # per-rank shard of the data
+x = torch.randn(64, 128, device=device)
+y = torch.randint(0, 32, (64,), device=device)
 for step in range(20):
-    x = torch.randn(64, 128, device=device)
-    y = torch.randint(0, 32, (64,), device=device)
     loss = torch.nn.functional.cross_entropy(net(x), y)
     zopt.zero_grad()
     loss.backward()

@@ -54,6 +54,7 @@ static auto communicator_class =
         .def("Alltoallv", &Communicator::Alltoallv)
         .def("AlltoallPairwise", &Communicator::AlltoallPairwise)
         .def("Iallreduce", &Communicator::Iallreduce)
+        .def("Ireducescatter", &Communicator::Ireducescatter)
         .def("Isend", &Communicator::Isend)
         .def("Irecv", &Communicator::Irecv)
         .def("Wait", &Communicator::Wait)

@@ -1506,6 +1506,34 @@ std::vector<at::Tensor> Communicator::Iallreduce(const Tensor& input,
   return {meta, out, in};
 }
 
+std::vector<at::Tensor> Communicator::Ireducescatter(const Tensor& input,
+                                                     int64_t op) {
+  check_op(op);
+  TORCH_CHECK(!torch::autograd::compute_requires_grad(input),
+              "mpi4torch_amd: Ireducescatter does not support autograd");
+  at::AutoDispatchBelowADInplaceOrView guard;
+  DeviceStager stager(input);
+  auto in = stager.to_comm(input).contiguous().variable_data();
+  auto& tr = tr_for(in);
+  TORCH_CHECK(in.numel() % tr.size() == 0,
+              "Ireducescatter: numel must be divisible by world size");
+  TORCH_CHECK(is_arith(op) && native_reduce_dtype(tr, in.scalar_type()),
+              "mpi4torch_amd: Ireducescatter supports native arithmetic "
+              "reductions only");
+  auto out = at::empty({in.numel() / tr.size()}, in.options());
+  uint64_t req = tr.ireduce_scatter(in, out, (RedOp)op);
+  auto meta = at::empty({7}, at::TensorOptions().dtype(at::kDouble));
+  auto* m = meta.data_ptr<double>();
+  m[0] = (double)req;
+  m[1] = (double)kIallreduceOp;  // same wait semantics: handle returns buffer
+  m[2] = 0.0;
+  m[3] = 0.0;
+  m[4] = ptr_hash(out.data_ptr());
+  m[5] = (double)(int64_t)stager.orig_device_.type();
+  m[6] = (double)stager.orig_device_.index();
+  return {meta, out, in};
+}
+
 std::vector<at::Tensor> Communicator::Isend(const Tensor& input, int64_t dest,
                                             int64_t tag) {
   return IsendImpl(input, dest, tag, /*backward_channel=*/false);

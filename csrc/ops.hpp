@@ -71,6 +71,10 @@ struct Communicator : torch::CustomClassHolder {
   // Allreduce for differentiable paths. Gradient bucketing (parallel/ddp)
   // is its main consumer.
   std::vector<at::Tensor> Iallreduce(const at::Tensor& input, int64_t op);
+  // Non-blocking equal-count reduce-scatter (no autograd): this rank's
+  // flat block of the elementwise sum. input.numel() must be
+  // world_size * block; returns a wait handle for the [block] result.
+  std::vector<at::Tensor> Ireducescatter(const at::Tensor& input, int64_t op);
 
   // Non-blocking p2p. Handle contract identical to the reference
   // (csrc/extension.cpp:1094-1107): [meta tensor, comm buffer, input].

@@ -337,6 +337,10 @@ struct LocalTransport final : Transport {
     TORCH_CHECK(peer == 0, "world_size is 1; cannot receive from rank ", peer);
     return self_matcher().irecv(buf, tag, ch);
   }
+  uint64_t ireduce_scatter(const at::Tensor& in, at::Tensor& out,
+                           RedOp op) override {
+    return iallreduce(in.view_as(out), out, op);
+  }
   uint64_t iallreduce(const at::Tensor& in, at::Tensor& out,
                       RedOp) override {
     out.copy_(in, true);
@@ -478,6 +482,15 @@ struct C10dTransport final : Transport {
     Request r;
     r.buffer = out;
     r.work = backend_->allreduce(ts, opts);
+    return requests().add(std::move(r));
+  }
+  uint64_t ireduce_scatter(const at::Tensor& in, at::Tensor& out,
+                           RedOp op) override {
+    // gloo has no reduce_scatter: synchronous emulation, pre-completed
+    // request (CPU correctness path; overlap is a GPU concern)
+    reduce_scatter_equal(in, out, op);
+    Request r;
+    r.buffer = out;
     return requests().add(std::move(r));
   }
 
@@ -647,6 +660,19 @@ struct RcclTransport final : Transport {
                             comm(ch), stream(ch)));
     return make_gpu_request(ch, buf);
   }
+  uint64_t ireduce_scatter(const at::Tensor& in, at::Tensor& out,
+                           RedOp op) override {
+    std::lock_guard<std::mutex> g(mu_);
+    check_not_capturing("Ireducescatter");
+    enter_side(Channel::Coll, {in, out});
+    M4A_NCCL_CHECK(ncclReduceScatter(in.data_ptr(), out.data_ptr(),
+                                     out.numel(),
+                                     nccl_reduce_dtype(out.scalar_type()),
+                                     nccl_red_op(op), comm(Channel::Coll),
+                                     stream(Channel::Coll)));
+    return make_gpu_request(Channel::Coll, out);
+  }
+
   // Runtime probe: does this RCCL build reduce fp8 natively? One tiny
   // allreduce per dtype, issued symmetrically on all ranks (the op layer
   // reaches this on every rank at the same program point). Cached.

@@ -71,6 +71,10 @@ struct Transport {
   // bucketing — an MI355X-first extension beyond the reference's API).
   virtual uint64_t iallreduce(const at::Tensor& in, at::Tensor& out,
                               RedOp op) = 0;
+  // Non-blocking equal-count reduce-scatter (no autograd): ZeRO-2 bucket
+  // primitive. in.numel() == size * out.numel(), rank-major blocks.
+  virtual uint64_t ireduce_scatter(const at::Tensor& in, at::Tensor& out,
+                                   RedOp op) = 0;
   // Whether fp8 reductions run natively (RCCL probes at runtime; the cast
   // fallback is used otherwise). Non-GPU transports: false.
   virtual bool fp8_reduce_supported(at::ScalarType) { return false; }

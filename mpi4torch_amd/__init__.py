@@ -224,6 +224,12 @@ class MPI_Communicator:
         not part of the reference API."""
         return WaitHandle(self._comm.Iallreduce(tensor, op))
 
+    def Ireducescatter(self, tensor: torch.Tensor, op: int) -> WaitHandle:
+        """Non-blocking equal-count flat reduce-scatter (no autograd):
+        Wait() yields this rank's block of the elementwise reduction.
+        The ZeRO-2 bucket primitive; not in the reference API."""
+        return WaitHandle(self._comm.Ireducescatter(tensor, op))
+
     def Isend(self, tensor: torch.Tensor, dest: int, tag: int) -> WaitHandle:
         """Non-blocking send; complete with Wait. Backward: reverse recv."""
         return WaitHandle(self._comm.Isend(tensor, dest, tag))

@@ -12,6 +12,7 @@ from .tp import (ColumnParallelLinear, RowParallelLinear, TensorParallelMLP,
                  copy_to_tp_region)
 from .pipeline import GPipe
 from .zero import ZeroRedundancyOptimizer
+from .sharded_ddp import ShardedDataParallel
 
 __all__ = [
     "DistributedDataParallel",
@@ -24,4 +25,5 @@ __all__ = [
     "copy_to_tp_region",
     "GPipe",
     "ZeroRedundancyOptimizer",
+    "ShardedDataParallel",
 ]

@@ -1,0 +1,137 @@
+"""ZeRO-2 data parallelism: bucketed gradient reduce-scatter overlapped
+with backward + sharded optimizer + parameter allgather.
+
+Per step and bucket, the wire carries exactly what DDP's allreduce would
+(reduce-scatter + allgather = one allreduce), but each rank keeps only
+1/P of the gradients (full per-bucket gradients are freed as soon as the
+bucket's reduce-scatter is enqueued) and 1/P of the optimizer state.
+
+Usage::
+
+    model = ShardedDataParallel(module, torch.optim.AdamW, lr=1e-3)
+    loss = model(x).sum()
+    loss.backward()          # bucket reduce-scatters overlap backward
+    model.step()             # wait -> sharded optimizer -> allgather params
+"""
+
+from typing import List
+
+import torch
+
+import mpi4torch_amd as m4a
+
+
+class _Bucket:
+    def __init__(self, params: List[torch.nn.Parameter], shard_len: int):
+        self.params = params
+        self.numels = [p.numel() for p in params]
+        self.shard_len = shard_len  # padded_len // P
+        self.pending = 0
+        self.handle = None
+        self.shard: torch.nn.Parameter = None  # this rank's slice
+
+
+class ShardedDataParallel(torch.nn.Module):
+    def __init__(self, module: torch.nn.Module, optimizer_cls, comm=None,
+                 bucket_cap_mb: int = 64, average: bool = True,
+                 **optim_kwargs):
+        super().__init__()
+        self.module = module
+        self.comm = comm if comm is not None else m4a.COMM_WORLD
+        self.average = average
+        P = self.comm.size
+
+        if P > 1:
+            with torch.no_grad():
+                for p in self.module.parameters():
+                    self.comm.Bcast_(p.data, 0)
+
+        params = [p for p in self.module.parameters() if p.requires_grad]
+        assert params, "no trainable parameters"
+        cap = bucket_cap_mb * 1024 * 1024
+        self._buckets: List[_Bucket] = []
+        cur: List[torch.nn.Parameter] = []
+        size = 0
+        for p in reversed(params):  # roughly backward completion order
+            cur.append(p)
+            size += p.numel() * p.element_size()
+            if size >= cap:
+                self._buckets.append(self._make_bucket(cur))
+                cur, size = [], 0
+        if cur:
+            self._buckets.append(self._make_bucket(cur))
+
+        self._param_bucket = {}
+        for b in self._buckets:
+            for p in b.params:
+                self._param_bucket[p] = b
+        for p in params:
+            p.register_post_accumulate_grad_hook(self._grad_ready)
+        self._reset_pending()
+
+        self.optimizer = optimizer_cls([b.shard for b in self._buckets],
+                                       **optim_kwargs)
+
+    def _make_bucket(self, params: List[torch.nn.Parameter]) -> _Bucket:
+        P = self.comm.size
+        total = sum(p.numel() for p in params)
+        shard_len = (total + P - 1) // P
+        b = _Bucket(list(params), shard_len)
+        # initialize this rank's parameter shard from the replicated params
+        with torch.no_grad():
+            flat = torch.zeros(shard_len * P, dtype=params[0].dtype,
+                               device=params[0].device)
+            torch.cat([p.reshape(-1) for p in params], out=flat[:total])
+            lo = self.comm.rank * shard_len
+            b.shard = flat[lo:lo + shard_len].clone().requires_grad_()
+        return b
+
+    def _reset_pending(self):
+        for b in self._buckets:
+            b.pending = len(b.params)
+            b.handle = None
+
+    def _grad_ready(self, p: torch.nn.Parameter):
+        b = self._param_bucket[p]
+        b.pending -= 1
+        if b.pending == 0:
+            with torch.no_grad():
+                P = self.comm.size
+                total = sum(b.numels)
+                flat = torch.zeros(b.shard_len * P, dtype=p.dtype,
+                                   device=p.device)
+                torch.cat([q.grad.reshape(-1) for q in b.params],
+                          out=flat[:total])
+                for q in b.params:
+                    q.grad = None  # ZeRO-2: full grads freed immediately
+                b.handle = (self.comm.Ireducescatter(flat, m4a.MPI_SUM)
+                            if P > 1 else None)
+                if P == 1:
+                    b.shard.grad = flat
+
+    def forward(self, *args, **kwargs):
+        return self.module(*args, **kwargs)
+
+    @torch.no_grad()
+    def step(self):
+        """Wait bucket reductions, run the sharded optimizer, allgather the
+        updated shards back into the module parameters."""
+        P = self.comm.size
+        for b in self._buckets:
+            if b.handle is not None:
+                g = self.comm.Wait(b.handle)
+                b.shard.grad = g / P if self.average else g
+        self.optimizer.step()
+        for b in self._buckets:
+            b.shard.grad = None
+            full = (self.comm.Allgather(b.shard.detach(), 0)
+                    if P > 1 else b.shard.detach())
+            off = 0
+            for q, n in zip(b.params, b.numels):
+                q.data.copy_(full[off:off + n].view_as(q))
+                off += n
+        self._reset_pending()
+
+    def zero_grad(self, set_to_none: bool = True):
+        for p in self.module.parameters():
+            p.grad = None

@@ -358,3 +358,12 @@ def test_gpu_full_path_fuzz(world1):
                 "pairwise", it, shape, axis, dtype)
     finally:
         m._C.force_full_path(False)
+
+
+def test_ireducescatter_gpu(world1):
+    m = world1
+    comm = m.COMM_WORLD
+    t = torch.rand(1 << 16, device="cuda")
+    h = comm.Ireducescatter(t, m.MPI_SUM)
+    out = comm.Wait(h)
+    torch.testing.assert_close(out, t)  # world of one

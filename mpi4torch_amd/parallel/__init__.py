@@ -13,6 +13,7 @@ from .tp import (ColumnParallelLinear, RowParallelLinear, TensorParallelMLP,
 from .pipeline import GPipe
 from .zero import ZeroRedundancyOptimizer
 from .sharded_ddp import ShardedDataParallel
+from .fsdp import FullyShardedDataParallel
 
 __all__ = [
     "DistributedDataParallel",
@@ -26,4 +27,5 @@ __all__ = [
     "GPipe",
     "ZeroRedundancyOptimizer",
     "ShardedDataParallel",
+    "FullyShardedDataParallel",
 ]

@@ -1,0 +1,177 @@
+"""ZeRO-3-lite (FSDP): parameters sharded at rest, materialized per unit.
+
+Each wrapped unit's parameters live as views into one flat buffer whose
+storage is freed (resized to 0) whenever the unit is idle; every rank
+keeps only a 1/P shard. Around a unit's forward and backward the full
+buffer is re-materialized with one Allgather; after backward the unit's
+gradients leave as one reduce-scatter and everything full-sized is freed
+again. Peak parameter memory is one unit, not the model.
+
+The storage-resize trick mirrors torch FSDP: autograd's saved tensors are
+views of the SAME storage, so restoring the storage before backward
+revalidates them in place.
+
+Usage::
+
+    model = FullyShardedDataParallel(module, units=[m1, m2, ...])
+    opt = torch.optim.AdamW(model.shard_parameters(), lr=...)
+    loss = model(x).sum(); loss.backward()
+    model.finish_backward()      # waits grad reduce-scatters
+    opt.step(); model.refresh_shards()
+"""
+
+from typing import List, Optional, Sequence
+
+import torch
+
+import mpi4torch_amd as m4a
+
+
+class _Unit:
+    def __init__(self, module: torch.nn.Module, comm):
+        self.module = module
+        self.comm = comm
+        self.params: List[torch.nn.Parameter] = [
+            p for p in module.parameters() if p.requires_grad
+        ]
+        assert self.params, "FSDP unit has no trainable parameters"
+        P = comm.size
+        self.numels = [p.numel() for p in self.params]
+        total = sum(self.numels)
+        self.shard_len = (total + P - 1) // P
+        self.padded = self.shard_len * P
+
+        p0 = self.params[0]
+        with torch.no_grad():
+            flat = torch.zeros(self.padded, dtype=p0.dtype, device=p0.device)
+            torch.cat([p.reshape(-1) for p in self.params], out=flat[:total])
+            lo = comm.rank * self.shard_len
+            # persistent 1/P shard (the optimizer's parameter)
+            self.shard = torch.nn.Parameter(
+                flat[lo:lo + self.shard_len].clone())
+            # the full buffer the module computes with; params become views
+            self.flat = flat
+            off = 0
+            for p, n in zip(self.params, self.numels):
+                p.data = flat[off:off + n].view_as(p)
+                off += n
+        self.materialized = True
+        self.grad_handle = None
+        self.pending = 0
+
+    @torch.no_grad()
+    def free(self):
+        if self.materialized:
+            self.flat.untyped_storage().resize_(0)
+            self.materialized = False
+
+    @torch.no_grad()
+    def materialize(self):
+        if self.materialized:
+            return
+        self.flat.untyped_storage().resize_(
+            self.padded * self.flat.element_size())
+        if self.comm.size > 1:
+            full = self.comm.Allgather(self.shard.detach(), 0)
+            self.flat.copy_(full)
+        else:
+            self.flat[:self.shard_len].copy_(self.shard.detach())
+        self.materialized = True
+
+    @torch.no_grad()
+    def start_grad_reduce(self):
+        total = sum(self.numels)
+        gflat = torch.zeros(self.padded, dtype=self.flat.dtype,
+                            device=self.flat.device)
+        off = 0
+        for p, n in zip(self.params, self.numels):
+            if p.grad is not None:
+                gflat[off:off + n].copy_(p.grad.reshape(-1))
+            p.grad = None
+            off += n
+        if self.comm.size > 1:
+            self.grad_handle = self.comm.Ireducescatter(gflat, m4a.MPI_SUM)
+        else:
+            self.shard.grad = gflat[:self.shard_len].clone()
+        self.free()  # full params not needed past this unit's backward
+
+    @torch.no_grad()
+    def finish_grad_reduce(self, average: bool):
+        if self.grad_handle is not None:
+            g = self.comm.Wait(self.grad_handle)
+            self.shard.grad = g / self.comm.size if average else g
+            self.grad_handle = None
+
+
+class FullyShardedDataParallel(torch.nn.Module):
+    def __init__(self, module: torch.nn.Module,
+                 units: Optional[Sequence[torch.nn.Module]] = None,
+                 comm=None, average: bool = True):
+        super().__init__()
+        self.module = module
+        self.comm = comm if comm is not None else m4a.COMM_WORLD
+        self.average = average
+        if self.comm.size > 1:
+            with torch.no_grad():
+                for p in module.parameters():
+                    self.comm.Bcast_(p.data, 0)
+        unit_modules = list(units) if units is not None else [
+            m for m in module.children()
+            if any(p.requires_grad for p in m.parameters())
+        ]
+        assert unit_modules, "no FSDP units found"
+        self._units = [_Unit(m, self.comm) for m in unit_modules]
+        self._by_module = {u.module: u for u in self._units}
+        self._by_param = {p: u for u in self._units for p in u.params}
+
+        for u in self._units:
+            u.module.register_forward_pre_hook(self._pre_forward)
+            u.module.register_forward_hook(self._post_forward)
+            u.module.register_full_backward_pre_hook(self._pre_backward)
+            for p in u.params:
+                p.register_post_accumulate_grad_hook(self._grad_ready)
+        # params at rest are sharded
+        for u in self._units:
+            u.free()
+
+    # ---- hooks -----------------------------------------------------------
+    def _pre_forward(self, module, inputs):
+        self._by_module[module].materialize()
+
+    def _post_forward(self, module, inputs, output):
+        u = self._by_module[module]
+        u.pending = len(u.params)
+        u.free()  # re-materialized by the pre-backward hook when training
+        return output
+
+    def _pre_backward(self, module, grad_output):
+        self._by_module[module].materialize()
+
+    def _grad_ready(self, p):
+        u = self._by_param[p]
+        u.pending -= 1
+        if u.pending == 0:
+            u.start_grad_reduce()
+
+    # ---- public ----------------------------------------------------------
+    def forward(self, *args, **kwargs):
+        return self.module(*args, **kwargs)
+
+    def shard_parameters(self):
+        return [u.shard for u in self._units]
+
+    def finish_backward(self):
+        """Wait all gradient reduce-scatters (call after loss.backward())."""
+        for u in self._units:
+            u.finish_grad_reduce(self.average)
+
+    @torch.no_grad()
+    def refresh_shards(self):
+        """No-op placeholder for symmetry: the optimizer updates the shards
+        in place; full parameters re-materialize lazily at next use."""
+        for u in self._units:
+            u.shard.grad = None
+
+    def zero_grad(self, set_to_none: bool = True):
+        for p in self.module.parameters():
+            p.grad = None

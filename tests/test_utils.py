@@ -66,3 +66,31 @@ def test_checkpoint_ws2(tmp_path_factory):
 
     d = tempfile.mkdtemp()
     run_spmd(2, _ckpt_worker, d)
+
+
+def _clip_worker(rank, world):
+    import torch
+    import mpi4torch_amd as m
+    from mpi4torch_amd.utils import clip_grad_norm_sharded
+
+    # shard: rank r holds a grad of known norm; global norm = sqrt(sum)
+    g = torch.full((4,), float(rank + 1))
+    p = torch.nn.Parameter(torch.zeros(4))
+    p.grad = g.clone()
+    total = clip_grad_norm_sharded([p], max_norm=1.0)
+    import math
+
+    expect = math.sqrt(sum(4 * (r + 1) ** 2 for r in range(world)))
+    assert abs(float(total) - expect) < 1e-6
+    # clipped to global norm 1: every slice scaled by 1/expect
+    assert torch.allclose(p.grad, g / expect, atol=1e-6)
+
+    # under the norm: untouched
+    p2 = torch.nn.Parameter(torch.zeros(2))
+    p2.grad = torch.full((2,), 1e-4)
+    clip_grad_norm_sharded([p2], max_norm=10.0)
+    assert torch.allclose(p2.grad, torch.full((2,), 1e-4))
+
+
+def test_clip_sharded_ws3():
+    run_spmd(3, _clip_worker)

@@ -22,7 +22,14 @@ import mpi4torch_amd as m4a
 
 class ZeroRedundancyOptimizer:
     def __init__(self, params: Iterable[torch.nn.Parameter], optimizer_cls,
-                 comm=None, average: bool = True, **optim_kwargs):
+                 comm=None, average: bool = True, master_dtype=None,
+                 **optim_kwargs):
+        """`master_dtype` (e.g. torch.float32 for a bf16 model) keeps the
+        sharded optimizer state and master weights in a wider dtype: the
+        fp32-master mixed-precision recipe with the master copy itself
+        sharded P ways. Gradients are reduced in the MODEL dtype (cheap
+        wire), upcast for the update, and parameters are re-broadcast in
+        the model dtype."""
         self.comm = comm if comm is not None else m4a.COMM_WORLD
         self.params: List[torch.nn.Parameter] = [
             p for p in params if p.requires_grad
@@ -36,13 +43,16 @@ class ZeroRedundancyOptimizer:
         self._padded = self._shard_len * P
 
         dev = self.params[0].device
-        dt = self.params[0].dtype
+        self._model_dtype = self.params[0].dtype
+        self._master_dtype = master_dtype or self._model_dtype
         with torch.no_grad():
-            flat = torch.zeros(self._padded, device=dev, dtype=dt)
+            flat = torch.zeros(self._padded, device=dev,
+                               dtype=self._model_dtype)
             torch.cat([p.reshape(-1) for p in self.params],
                       out=flat[:total])
         lo = self.comm.rank * self._shard_len
-        self._shard = flat[lo:lo + self._shard_len].clone().requires_grad_()
+        self._shard = (flat[lo:lo + self._shard_len].clone()
+                       .to(self._master_dtype).requires_grad_())
         self.optimizer = optimizer_cls([self._shard], **optim_kwargs)
 
     def zero_grad(self, set_to_none: bool = True):
@@ -58,7 +68,7 @@ class ZeroRedundancyOptimizer:
         total = sum(self._numels)
         dev = self._shard.device
         gflat = torch.zeros(self._padded, device=dev,
-                            dtype=self._shard.dtype)
+                            dtype=self._model_dtype)
         off = 0
         for p, n in zip(self.params, self._numels):
             if p.grad is not None:
@@ -69,12 +79,12 @@ class ZeroRedundancyOptimizer:
         gshard = self.comm.Reducescatter(gflat, 0, self._shard_len)
         if self.average and P > 1:
             gshard = gshard / P
-        self._shard.grad = gshard
+        self._shard.grad = gshard.to(self._master_dtype)
         self.optimizer.step()
         self._shard.grad = None
-        # materialize updated parameters everywhere
-        full = (self.comm.Allgather(self._shard.detach(), 0)
-                if P > 1 else self._shard.detach())
+        # materialize updated parameters everywhere (model dtype on the wire)
+        local = self._shard.detach().to(self._model_dtype)
+        full = self.comm.Allgather(local, 0) if P > 1 else local
         off = 0
         for p, n in zip(self.params, self._numels):
             p.data.copy_(full[off:off + n].view_as(p))

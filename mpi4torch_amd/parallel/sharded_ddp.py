@@ -34,11 +34,12 @@ class _Bucket:
 class ShardedDataParallel(torch.nn.Module):
     def __init__(self, module: torch.nn.Module, optimizer_cls, comm=None,
                  bucket_cap_mb: int = 64, average: bool = True,
-                 **optim_kwargs):
+                 master_dtype=None, **optim_kwargs):
         super().__init__()
         self.module = module
         self.comm = comm if comm is not None else m4a.COMM_WORLD
         self.average = average
+        self._master_dtype = master_dtype  # None = model dtype
         P = self.comm.size
 
         if P > 1:
@@ -83,7 +84,10 @@ class ShardedDataParallel(torch.nn.Module):
                                device=params[0].device)
             torch.cat([p.reshape(-1) for p in params], out=flat[:total])
             lo = self.comm.rank * shard_len
-            b.shard = flat[lo:lo + shard_len].clone().requires_grad_()
+            shard = flat[lo:lo + shard_len].clone()
+            if self._master_dtype is not None:
+                shard = shard.to(self._master_dtype)  # fp32 master shard
+            b.shard = shard.requires_grad_()
         return b
 
     def _reset_pending(self):
@@ -107,7 +111,7 @@ class ShardedDataParallel(torch.nn.Module):
                 b.handle = (self.comm.Ireducescatter(flat, m4a.MPI_SUM)
                             if P > 1 else None)
                 if P == 1:
-                    b.shard.grad = flat
+                    b.shard.grad = flat.to(b.shard.dtype)
 
     def forward(self, *args, **kwargs):
         return self.module(*args, **kwargs)
@@ -120,12 +124,17 @@ class ShardedDataParallel(torch.nn.Module):
         for b in self._buckets:
             if b.handle is not None:
                 g = self.comm.Wait(b.handle)
-                b.shard.grad = g / P if self.average else g
+                if self.average:
+                    g = g / P
+                b.shard.grad = (g.to(b.shard.dtype)
+                                if g.dtype != b.shard.dtype else g)
         self.optimizer.step()
         for b in self._buckets:
             b.shard.grad = None
-            full = (self.comm.Allgather(b.shard.detach(), 0)
-                    if P > 1 else b.shard.detach())
+            local = b.shard.detach()
+            if local.dtype != b.params[0].dtype:
+                local = local.to(b.params[0].dtype)  # model dtype on the wire
+            full = self.comm.Allgather(local, 0) if P > 1 else local
             off = 0
             for q, n in zip(b.params, b.numels):
                 q.data.copy_(full[off:off + n].view_as(q))

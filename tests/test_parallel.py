@@ -359,3 +359,44 @@ def _zero_mixed_worker(rank, world):
 
 def test_zero_mixed_precision_ws2():
     run_spmd(2, _zero_mixed_worker)
+
+
+def _zero2_mixed_worker(rank, world):
+    import mpi4torch_amd as m
+    from mpi4torch_amd.parallel import ShardedDataParallel
+
+    torch.manual_seed(91)
+    net = torch.nn.Linear(8, 4).to(torch.bfloat16)
+    model = ShardedDataParallel(net, torch.optim.SGD, bucket_cap_mb=1,
+                                master_dtype=torch.float32, lr=0.5)
+    comm = m.COMM_WORLD
+    master = {n: p.detach().float().clone()
+              for n, p in net.named_parameters()}
+    for step in range(6):
+        torch.manual_seed(5 * step + rank)
+        x = torch.randn(4, 8).to(torch.bfloat16)
+        loss = model(x).square().sum()
+        model.zero_grad()
+        # reference grads BEFORE hooks clear them: use a clone net? simpler:
+        # recompute on a replica
+        import copy
+
+        rep = copy.deepcopy(net)
+        ref_loss = rep(x).square().sum()
+        ref_loss.backward()
+        grads = {n: (comm.Allreduce(p.grad, m.MPI_SUM).float() / world)
+                 for n, p in rep.named_parameters()}
+        loss.backward()
+        model.step()
+        with torch.no_grad():
+            for n, p in net.named_parameters():
+                master[n] -= 0.5 * grads[n]
+                assert torch.equal(p.detach(),
+                                   master[n].to(torch.bfloat16)), (
+                    "ZeRO-2 mixed master diverged", n, step)
+    for b in model._buckets:
+        assert b.shard.dtype == torch.float32
+
+
+def test_zero2_mixed_ws2():
+    run_spmd(2, _zero2_mixed_worker)

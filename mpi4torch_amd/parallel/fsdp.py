@@ -28,7 +28,7 @@ import mpi4torch_amd as m4a
 
 
 class _Unit:
-    def __init__(self, module: torch.nn.Module, comm):
+    def __init__(self, module: torch.nn.Module, comm, master_dtype=None):
         self.module = module
         self.comm = comm
         self.params: List[torch.nn.Parameter] = [
@@ -46,9 +46,12 @@ class _Unit:
             flat = torch.zeros(self.padded, dtype=p0.dtype, device=p0.device)
             torch.cat([p.reshape(-1) for p in self.params], out=flat[:total])
             lo = comm.rank * self.shard_len
-            # persistent 1/P shard (the optimizer's parameter)
-            self.shard = torch.nn.Parameter(
-                flat[lo:lo + self.shard_len].clone())
+            # persistent 1/P shard (the optimizer's parameter); optionally a
+            # wider master dtype (fp32 masters for a bf16 model)
+            shard = flat[lo:lo + self.shard_len].clone()
+            if master_dtype is not None:
+                shard = shard.to(master_dtype)
+            self.shard = torch.nn.Parameter(shard)
             # the full buffer the module computes with; params become views
             self.flat = flat
             off = 0
@@ -71,11 +74,14 @@ class _Unit:
             return
         self.flat.untyped_storage().resize_(
             self.padded * self.flat.element_size())
+        local = self.shard.detach()
+        if local.dtype != self.flat.dtype:
+            local = local.to(self.flat.dtype)  # model dtype on the wire
         if self.comm.size > 1:
-            full = self.comm.Allgather(self.shard.detach(), 0)
+            full = self.comm.Allgather(local, 0)
             self.flat.copy_(full)
         else:
-            self.flat[:self.shard_len].copy_(self.shard.detach())
+            self.flat[:self.shard_len].copy_(local)
         self.materialized = True
 
     @torch.no_grad()
@@ -92,21 +98,25 @@ class _Unit:
         if self.comm.size > 1:
             self.grad_handle = self.comm.Ireducescatter(gflat, m4a.MPI_SUM)
         else:
-            self.shard.grad = gflat[:self.shard_len].clone()
+            self.shard.grad = (gflat[:self.shard_len].clone()
+                               .to(self.shard.dtype))
         self.free()  # full params not needed past this unit's backward
 
     @torch.no_grad()
     def finish_grad_reduce(self, average: bool):
         if self.grad_handle is not None:
             g = self.comm.Wait(self.grad_handle)
-            self.shard.grad = g / self.comm.size if average else g
+            if average:
+                g = g / self.comm.size
+            self.shard.grad = (g.to(self.shard.dtype)
+                               if g.dtype != self.shard.dtype else g)
             self.grad_handle = None
 
 
 class FullyShardedDataParallel(torch.nn.Module):
     def __init__(self, module: torch.nn.Module,
                  units: Optional[Sequence[torch.nn.Module]] = None,
-                 comm=None, average: bool = True):
+                 comm=None, average: bool = True, master_dtype=None):
         super().__init__()
         self.module = module
         self.comm = comm if comm is not None else m4a.COMM_WORLD
@@ -120,7 +130,8 @@ class FullyShardedDataParallel(torch.nn.Module):
             if any(p.requires_grad for p in m.parameters())
         ]
         assert unit_modules, "no FSDP units found"
-        self._units = [_Unit(m, self.comm) for m in unit_modules]
+        self._units = [_Unit(m, self.comm, master_dtype)
+                       for m in unit_modules]
         self._by_module = {u.module: u for u in self._units}
         self._by_param = {p: u for u in self._units for p in u.params}
 

@@ -400,3 +400,31 @@ def _zero2_mixed_worker(rank, world):
 
 def test_zero2_mixed_ws2():
     run_spmd(2, _zero2_mixed_worker)
+
+
+def _fsdp_mixed_worker(rank, world):
+    from mpi4torch_amd.parallel.fsdp import FullyShardedDataParallel
+
+    torch.manual_seed(14)
+    net = torch.nn.Sequential(torch.nn.Linear(6, 6)).to(torch.bfloat16)
+    model = FullyShardedDataParallel(net, units=[net[0]],
+                                     master_dtype=torch.float32)
+    opt = torch.optim.SGD(model.shard_parameters(), lr=0.25)
+    for u in model._units:
+        assert u.shard.dtype == torch.float32
+    for step in range(3):
+        torch.manual_seed(3 * step + rank)
+        x = torch.randn(4, 6).to(torch.bfloat16)
+        loss = model(x).square().sum()
+        model.zero_grad()
+        loss.backward()
+        model.finish_backward()
+        opt.step()
+        model.refresh_shards()
+    for u in model._units:
+        u.materialize()
+    assert net[0].weight.dtype == torch.bfloat16
+
+
+def test_fsdp_mixed_ws2():
+    run_spmd(2, _fsdp_mixed_worker)

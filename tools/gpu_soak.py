@@ -30,7 +30,9 @@ def main():
     if not dist.is_initialized():
         dist.init_process_group("gloo", rank=0, world_size=1)
     import mpi4torch_amd as m
-    from mpi4torch_amd.parallel import DistributedDataParallel
+    from mpi4torch_amd.models.moe import ExpertParallelMoE
+    from mpi4torch_amd.parallel import (DistributedDataParallel,
+                                        ShardedDataParallel)
 
     m.init()
     comm = m.COMM_WORLD
@@ -41,6 +43,13 @@ def main():
     ).to(device)
     model = DistributedDataParallel(net, bucket_cap_mb=4)
     opt = torch.optim.AdamW(model.parameters(), lr=1e-4)
+    znet = torch.nn.Sequential(
+        torch.nn.Linear(128, 512), torch.nn.GELU(), torch.nn.Linear(512, 32)
+    ).to(device)
+    zmodel = ShardedDataParallel(znet, torch.optim.AdamW, bucket_cap_mb=1,
+                                 lr=1e-4)
+    moe = ExpertParallelMoE(64, 4).to(device)
+    moe_opt = torch.optim.AdamW(moe.parameters(), lr=1e-4)
 
     m._C.force_full_path(True)
     torch.cuda.synchronize()
@@ -56,6 +65,16 @@ def main():
         loss.backward()
         model.finish_gradient_sync()
         opt.step()
+
+        # ZeRO-2 step + expert-parallel MoE step
+        zloss = zmodel(torch.randn(32, 128, device=device)).square().mean()
+        zmodel.zero_grad()
+        zloss.backward()
+        zmodel.step()
+        mloss = moe(torch.randn(48, 64, device=device)).square().mean()
+        moe_opt.zero_grad()
+        mloss.backward()
+        moe_opt.step()
 
         # axis collectives, full pipeline, with value checks
         t = torch.rand(4, 1000 + (it % 7), 8, device=device)

@@ -121,6 +121,7 @@ class FullyShardedDataParallel(torch.nn.Module):
         self.module = module
         self.comm = comm if comm is not None else m4a.COMM_WORLD
         self.average = average
+        self._sync_enabled = True
         if self.comm.size > 1:
             with torch.no_grad():
                 for p in module.parameters():
@@ -159,6 +160,8 @@ class FullyShardedDataParallel(torch.nn.Module):
         self._by_module[module].materialize()
 
     def _grad_ready(self, p):
+        if not self._sync_enabled:
+            return  # accumulation: keep full local grads; unit stays freed
         u = self._by_param[p]
         u.pending -= 1
         if u.pending == 0:
@@ -182,6 +185,21 @@ class FullyShardedDataParallel(torch.nn.Module):
         in place; full parameters re-materialize lazily at next use."""
         for u in self._units:
             u.shard.grad = None
+
+    def no_sync(self):
+        """Gradient-accumulation context: local .grad accumulates; the
+        final backward outside the context reduces the sums. (Parameters
+        still materialize/free around each microbatch's forward/backward.)"""
+        f = self
+
+        class _NoSync:
+            def __enter__(self):
+                f._sync_enabled = False
+
+            def __exit__(self, *exc):
+                f._sync_enabled = True
+
+        return _NoSync()
 
     def zero_grad(self, set_to_none: bool = True):
         for p in self.module.parameters():

@@ -40,6 +40,7 @@ class ShardedDataParallel(torch.nn.Module):
         self.comm = comm if comm is not None else m4a.COMM_WORLD
         self.average = average
         self._master_dtype = master_dtype  # None = model dtype
+        self._sync_enabled = True
         P = self.comm.size
 
         if P > 1:
@@ -96,6 +97,8 @@ class ShardedDataParallel(torch.nn.Module):
             b.handle = None
 
     def _grad_ready(self, p: torch.nn.Parameter):
+        if not self._sync_enabled:
+            return  # accumulation microbatch: keep local grads
         b = self._param_bucket[p]
         b.pending -= 1
         if b.pending == 0:
@@ -140,6 +143,21 @@ class ShardedDataParallel(torch.nn.Module):
                 q.data.copy_(full[off:off + n].view_as(q))
                 off += n
         self._reset_pending()
+
+    def no_sync(self):
+        """Context manager for gradient-accumulation microbatches: local
+        gradients accumulate in .grad; the final backward OUTSIDE the
+        context reduces the accumulated values."""
+        sdp = self
+
+        class _NoSync:
+            def __enter__(self):
+                sdp._sync_enabled = False
+
+            def __exit__(self, *exc):
+                sdp._sync_enabled = True
+
+        return _NoSync()
 
     def zero_grad(self, set_to_none: bool = True):
         for p in self.module.parameters():

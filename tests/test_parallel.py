@@ -428,3 +428,84 @@ def _fsdp_mixed_worker(rank, world):
 
 def test_fsdp_mixed_ws2():
     run_spmd(2, _fsdp_mixed_worker)
+
+
+def _zero2_accum_worker(rank, world):
+    import copy
+
+    import mpi4torch_amd as m
+    from mpi4torch_amd.parallel import ShardedDataParallel
+
+    torch.manual_seed(17)
+    net = torch.nn.Linear(6, 3, bias=False).double()
+    model = ShardedDataParallel(net, torch.optim.SGD, bucket_cap_mb=1,
+                                lr=0.1)
+    ref_net = copy.deepcopy(net)
+    ref_opt = torch.optim.SGD(ref_net.parameters(), lr=0.1)
+    comm = m.COMM_WORLD
+
+    torch.manual_seed(33 + rank)
+    xs = [torch.randn(4, 6, dtype=torch.double) for _ in range(3)]
+    # 2 accumulation microbatches + 1 syncing one
+    with model.no_sync():
+        model(xs[0]).square().sum().backward()
+        model(xs[1]).square().sum().backward()
+    model(xs[2]).square().sum().backward()
+    model.step()
+
+    for x in xs:
+        ref_net(x).square().sum().backward()
+    with torch.no_grad():
+        for p in ref_net.parameters():
+            p.grad.copy_(comm.Allreduce(p.grad, m.MPI_SUM) / world)
+    ref_opt.step()
+
+    for p, q in zip(net.parameters(), ref_net.parameters()):
+        assert torch.allclose(p, q, atol=1e-12), (
+            "accumulated ZeRO-2 diverged", (p - q).abs().max())
+
+
+def _fsdp_accum_worker(rank, world):
+    import copy
+
+    import mpi4torch_amd as m
+    from mpi4torch_amd.parallel.fsdp import FullyShardedDataParallel
+
+    comm = m.COMM_WORLD
+    torch.manual_seed(19)
+    net = torch.nn.Sequential(torch.nn.Linear(5, 5, bias=False)).double()
+    model = FullyShardedDataParallel(net, units=[net[0]])
+    opt = torch.optim.SGD(model.shard_parameters(), lr=0.1)
+    torch.manual_seed(19)
+    ref_net = torch.nn.Sequential(torch.nn.Linear(5, 5, bias=False)).double()
+    ref_opt = torch.optim.SGD(ref_net.parameters(), lr=0.1)
+
+    torch.manual_seed(40 + rank)
+    xs = [torch.randn(3, 5, dtype=torch.double) for _ in range(2)]
+    with model.no_sync():
+        model(xs[0]).square().sum().backward()
+    model(xs[1]).square().sum().backward()
+    model.finish_backward()
+    opt.step()
+    model.refresh_shards()
+
+    for x in xs:
+        ref_net(x).square().sum().backward()
+    with torch.no_grad():
+        for p in ref_net.parameters():
+            p.grad.copy_(comm.Allreduce(p.grad, m.MPI_SUM) / world)
+    ref_opt.step()
+
+    for u in model._units:
+        u.materialize()
+    for p, q in zip(net.parameters(), ref_net.parameters()):
+        assert torch.allclose(p, q, atol=1e-12), (
+            "accumulated FSDP diverged", (p - q).abs().max())
+
+
+def test_zero2_accumulation_ws2():
+    run_spmd(2, _zero2_accum_worker)
+
+
+def test_fsdp_accumulation_ws2():
+    run_spmd(2, _fsdp_accum_worker)

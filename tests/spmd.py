@@ -42,7 +42,7 @@ def run_spmd(world_size, fn, *args):
     just-freed rendezvous port (TOCTOU between _free_port and gloo's bind).
     """
     last = None
-    for _ in range(2):
+    for _ in range(3):
         port = _free_port()
         try:
             mp.spawn(

@@ -38,7 +38,7 @@ def _entry(rank, world_size, port, fn, args):
 def run_spmd(world_size, fn, *args):
     """Spawn `world_size` ranks, each running fn(rank, world_size, *args).
 
-    Retries once on failure: rapid-fire test sessions can race on the
+    Retries (3 attempts): rapid-fire test sessions can race on the
     just-freed rendezvous port (TOCTOU between _free_port and gloo's bind).
     """
     last = None

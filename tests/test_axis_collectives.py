@@ -339,3 +339,34 @@ def test_reducescatter_ws2():
 
 def test_reducescatter_ws5():
     run_spmd(5, _reducescatter_worker)
+
+
+def _extension_errors_worker(rank, world):
+    import mpi4torch_amd as m
+
+    comm = m.COMM_WORLD
+    t = torch.rand(2, world, 3, dtype=torch.double)
+
+    def expect_error(fn, needle):
+        try:
+            fn()
+            raise AssertionError(f"expected error containing '{needle}'")
+        except RuntimeError as e:
+            assert needle in str(e), (needle, str(e))
+
+    expect_error(
+        lambda: comm.Alltoallv(t, 0, 1, [1] * (world + 1), [2] * (world + 1)),
+        "world_size entries")
+    expect_error(
+        lambda: comm.Reducescatter(torch.rand(2, world + 3, 2), 1, 1),
+        "must equal the axis size")
+    expect_error(
+        lambda: comm.AlltoallPairwise(t, 1, [2] * world, []),
+        "must equal the axis size")
+    expect_error(
+        lambda: comm.AlltoallPairwise(t, 1, [1] * (world + 2), []),
+        "world_size entries")
+
+
+def test_extension_errors_ws2():
+    run_spmd(2, _extension_errors_worker)

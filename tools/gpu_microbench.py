@@ -96,6 +96,20 @@ def main():
         print(f"allreduce {mib} MiB bf16: {t*1e3:.3f} ms "
               f"{nbytes/t/1e9:.1f} GB/s algbw")
 
+    print("== small-op latency (per call, current-stream wall) ==")
+    for nbytes, label in ((4096, "4KB"), (1 << 20, "1MB")):
+        xs = torch.randn(nbytes // 4, device="cuda")
+        t = timed(lambda: comm.Allreduce(xs, m.MPI_SUM), iters=200, warmup=20)
+        print(f"allreduce {label}: {t*1e6:.1f} us/op")
+        h = None
+
+        def ia():
+            hh = comm.Iallreduce(xs, m.MPI_SUM)
+            comm.Wait(hh)
+
+        t = timed(ia, iters=200, warmup=20)
+        print(f"iallreduce+wait {label}: {t*1e6:.1f} us/op")
+
     print("== allreduce fwd+bwd step (bench inner loop) ==")
     x = torch.randn(1 << 29, device="cuda", dtype=torch.float32).to(
         torch.bfloat16).requires_grad_()

@@ -132,6 +132,15 @@ class MPI_Communicator:
         """Number of processes in this communicator."""
         return self._comm.GetSize()
 
+    @property
+    def group_name(self) -> str:
+        """The c10d group this communicator is bound to ('' = local)."""
+        return self._comm.GetGroupName()
+
+    def Barrier(self) -> None:
+        """Synchronize all ranks (host-level; not in the reference API)."""
+        self._comm.Allreduce(torch.zeros(1), MPI_SUM)
+
     def Allreduce(self, tensor: torch.Tensor, op: int) -> torch.Tensor:
         """Elementwise combine across all ranks; result on every rank.
 

@@ -139,7 +139,9 @@ class MPI_Communicator:
 
     def Barrier(self) -> None:
         """Synchronize all ranks (host-level; not in the reference API)."""
-        self._comm.Allreduce(torch.zeros(1), MPI_SUM)
+        # literal 2 == MPI_SUM (module globals are not visible inside a
+        # TorchScript class body)
+        self._comm.Allreduce(torch.zeros(1), 2)
 
     def Allreduce(self, tensor: torch.Tensor, op: int) -> torch.Tensor:
         """Elementwise combine across all ranks; result on every rank.

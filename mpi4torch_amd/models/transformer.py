@@ -17,7 +17,7 @@ in tests/test_models.py).
 import torch
 
 import mpi4torch_amd as m4a
-from mpi4torch_amd.parallel.ulysses import seq_to_head, head_to_seq
+from mpi4torch_amd.parallel.ulysses import head_to_seq
 
 
 class UlyssesTransformerBlock(torch.nn.Module):

@@ -152,6 +152,13 @@ def test_host_staging_toggle(world1):
         torch.testing.assert_close(r, t.detach())
         r.sum().backward()
         torch.testing.assert_close(t.grad, torch.ones_like(t))
+        # staged variants of the axis/extension ops
+        x = torch.rand(2, 6, 3, device="cuda")
+        torch.testing.assert_close(comm.Gather(x, 1, 0), x)
+        torch.testing.assert_close(comm.Alltoallv(x, 1, 1, [6], [6]), x)
+        torch.testing.assert_close(comm.Reducescatter(x, 1, 6), x)
+        torch.testing.assert_close(
+            comm.AlltoallPairwise(x, 1, [6], []), x)
     finally:
         m.force_host_staging(False)
 

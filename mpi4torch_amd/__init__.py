@@ -64,6 +64,7 @@ __all__ = [
     "COMM_WORLD",
     "init",
     "comm_from_process_group",
+    "comm_split",
     "comm_from_mpi4py",
     "force_host_staging",
     "deactivate_cuda_aware_mpi_support",
@@ -348,6 +349,37 @@ def comm_from_process_group(pg) -> MPI_Communicator:
         companion = dist.new_group(ranks=ranks, backend="gloo")
         name = companion.group_name
     return MPI_Communicator(torch.ops.mpi4torch_amd.comm_from_group(name))
+
+
+def comm_split(comm: MPI_Communicator, color: int) -> Optional[MPI_Communicator]:
+    """MPI_Comm_split analog: partition `comm` into sub-communicators by
+    `color`. Color < 0 (MPI_UNDEFINED style) returns None. Collective over
+    `comm` (every rank must call it).
+
+    Members are ranked by ascending global rank (torch.distributed groups
+    fix this ordering, so MPI's `key` reordering is not supported).
+
+    The reference only obtains sub-communicators through mpi4py
+    (src/__init__.py:247-261); here they are torch.distributed groups
+    created on the fly.
+    """
+    import torch.distributed as dist
+
+    world = comm.size
+    colors_t = comm.Allgather(
+        torch.tensor([float(color)], dtype=torch.float64), 0)
+    colors = [int(colors_t[r].item()) for r in range(world)]
+    mine = None
+    # every rank must create EVERY subgroup (new_group is collective over
+    # the default world)
+    for c in sorted(set(c for c in colors if c >= 0)):
+        ranks = [r for r in range(world) if colors[r] == c]
+        g = dist.new_group(ranks=ranks, backend="gloo")
+        if c == color:
+            mine = g
+    if color < 0 or mine is None:
+        return None
+    return comm_from_process_group(mine)
 
 
 def comm_from_mpi4py(comm) -> MPI_Communicator:

@@ -83,3 +83,29 @@ def test_subgroups_ws5():
 
 def test_pickle_ws2():
     run_spmd(2, _pickle_worker)
+
+
+def _split_worker(rank, world):
+    import mpi4torch_amd as m
+
+    comm = m.COMM_WORLD
+    # three colors; key reverses the order inside each color
+    color = rank % 3
+    sub = m.comm_split(comm, color)
+    members = sorted(r for r in range(world) if r % 3 == color)
+    assert sub.size == len(members)
+    assert sub.rank == members.index(rank), (rank, sub.rank)
+    t = torch.full((2,), float(rank), dtype=torch.double)
+    s = sub.Allreduce(t, m.MPI_SUM)
+    assert (s == sum(members)).all()
+
+    # undefined color drops out (but must still participate in the call)
+    sub2 = m.comm_split(comm, -1 if rank == 0 else 7)
+    if rank == 0:
+        assert sub2 is None
+    else:
+        assert sub2 is not None and sub2.size == world - 1
+
+
+def test_comm_split_ws5():
+    run_spmd(5, _split_worker)

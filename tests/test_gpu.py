@@ -438,3 +438,35 @@ def test_zero_family_gpu(world1):
         u.materialize()
     for p, q in zip(net3.parameters(), ref.parameters()):
         torch.testing.assert_close(p, q, rtol=0, atol=1e-12)
+
+
+def test_nondefault_stream_ordering(world1):
+    """Collectives issued from a user (non-default) stream must be ordered
+    by the event bracket: producer kernel -> collective -> consumer kernel
+    across three different streams, with no host syncs in between."""
+    m = world1
+    comm = m.COMM_WORLD
+    s1 = torch.cuda.Stream()
+    s2 = torch.cuda.Stream()
+    n = 1 << 20
+    with torch.cuda.stream(s1):
+        x = torch.ones(n, device="cuda")
+        x = x * 3  # producer on s1
+        r = comm.Allreduce(x, m.MPI_SUM)  # collective from s1
+    s2.wait_stream(s1)
+    with torch.cuda.stream(s2):
+        y = r + 1  # consumer on s2
+    torch.cuda.current_stream().wait_stream(s2)
+    torch.cuda.synchronize()
+    assert (y == 4).all()
+
+    # p2p from a side stream; Wait issued from the default stream
+    with torch.cuda.stream(s1):
+        t = torch.full((n,), 7.0, device="cuda")
+        h = comm.Isend(t, 0, 1)
+        h2 = comm.Irecv(torch.empty(n, device="cuda"), 0, 1)
+    torch.cuda.current_stream().wait_stream(s1)
+    got = comm.Wait(h2)
+    comm.Wait(h)
+    torch.cuda.synchronize()
+    assert (got == 7).all()

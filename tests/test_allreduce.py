@@ -162,3 +162,24 @@ def _double_backward_worker(rank, world):
 
 def test_double_backward_ws2():
     run_spmd(2, _double_backward_worker)
+
+
+def _threads_worker(rank, world):
+    import concurrent.futures as cf
+
+    import mpi4torch_amd as m
+
+    comm = m.COMM_WORLD
+    # NOTE: each thread issues a self-contained collective; ACROSS ranks the
+    # global order must still match, so every thread's op must be identical
+    # here (same shape/op) — we only check values and absence of deadlock.
+    def job(i):
+        t = torch.ones(512)
+        return bool((comm.Allreduce(t, m.MPI_SUM) == world).all())
+
+    with cf.ThreadPoolExecutor(max_workers=4) as ex:
+        assert all(ex.map(job, range(32)))
+
+
+def test_multithreaded_ws2():
+    run_spmd(2, _threads_worker)

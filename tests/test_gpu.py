@@ -470,3 +470,24 @@ def test_nondefault_stream_ordering(world1):
     comm.Wait(h)
     torch.cuda.synchronize()
     assert (got == 7).all()
+
+
+def test_multithreaded_collectives(world1):
+    """Concurrent collective issue from several host threads (the autograd
+    engine uses its own threads in real training): per-communicator mutex
+    + event pool must serialize enqueues without loss or deadlock."""
+    import concurrent.futures as cf
+
+    m = world1
+    comm = m.COMM_WORLD
+
+    def job(i):
+        t = torch.full((4096,), float(i), device="cuda")
+        r = comm.Allreduce(t, m.MPI_SUM)
+        g = comm.Gather(t, 0, 0)
+        torch.cuda.synchronize()
+        return bool((r == i).all()) and bool((g == i).all())
+
+    with cf.ThreadPoolExecutor(max_workers=4) as ex:
+        results = list(ex.map(job, range(64)))
+    assert all(results)

@@ -68,10 +68,23 @@ namespace {
 struct Request {
   bool gpu = false;
   bool self_pending = false;  // self-p2p not yet matched
+  // Deferred RCCL p2p: the nccl call has not been issued yet; wait_request
+  // triggers owner->flush_p2p(), which batches every pending send/recv of
+  // this transport into one ncclGroupStart/End (so matched pairs can
+  // rendezvous) and fills in `event`.
+  bool deferred = false;
+  std::weak_ptr<Transport> owner;
   hipEvent_t event = nullptr;
   int device = -1;
   c10::intrusive_ptr<c10d::Work> work;
   at::Tensor buffer;  // keeps the comm buffer alive until wait
+  // MPI4TORCH_AMD_DEBUG=1 p2p handshake: the sender ships [tag, dtype,
+  // numel] over the host channel; the receiver compares at Wait and raises
+  // on mismatch — FIFO-crossed transfers become errors, not silent swaps.
+  c10::intrusive_ptr<c10d::Work> debug_work;
+  at::Tensor debug_meta;
+  std::vector<int64_t> debug_expect;  // recv side: [tag, dtype, numel]
+  int debug_peer = -1;
 };
 
 struct RequestTable {
@@ -157,10 +170,22 @@ struct SelfMatcher {
   struct Pending {
     uint64_t req_id;
     at::Tensor buf;
+    int tag;
   };
   std::mutex mu;
-  // keyed by (channel, tag)
-  std::map<std::pair<int, int64_t>, std::deque<Pending>> sends, recvs;
+  // Keyed by channel only: matching is FIFO per (peer, channel) on EVERY
+  // transport (RCCL has no tags), so self traffic follows the same
+  // contract. Tags are validated at match time under MPI4TORCH_AMD_DEBUG=1.
+  std::map<int, std::deque<Pending>> sends, recvs;
+
+  static void check_tags(int stag, int rtag) {
+    if (!config().debug_collectives) return;
+    TORCH_CHECK(stag == rtag,
+                "mpi4torch_amd[debug]: self send/recv matched FIFO but with "
+                "different tags (send tag ", stag, ", recv tag ", rtag,
+                "). Matching is FIFO per (peer, channel); tags do not "
+                "reorder it — reorder your posts instead.");
+  }
 
   static void complete_pair(const at::Tensor& src, at::Tensor& dst,
                             uint64_t sreq, uint64_t rreq) {
@@ -187,15 +212,15 @@ struct SelfMatcher {
     r.self_pending = true;
     uint64_t id = requests().add(std::move(r));
     std::unique_lock<std::mutex> g(mu);
-    auto key = std::make_pair((int)ch, (int64_t)tag);
-    auto& rq = recvs[key];
+    auto& rq = recvs[(int)ch];
     if (!rq.empty()) {
       Pending p = rq.front();
       rq.pop_front();
       g.unlock();
+      check_tags(tag, p.tag);
       complete_pair(buf, p.buf, id, p.req_id);
     } else {
-      sends[key].push_back({id, buf});
+      sends[(int)ch].push_back({id, buf, tag});
     }
     return id;
   }
@@ -205,15 +230,15 @@ struct SelfMatcher {
     r.self_pending = true;
     uint64_t id = requests().add(std::move(r));
     std::unique_lock<std::mutex> g(mu);
-    auto key = std::make_pair((int)ch, (int64_t)tag);
-    auto& sq = sends[key];
+    auto& sq = sends[(int)ch];
     if (!sq.empty()) {
       Pending p = sq.front();
       sq.pop_front();
       g.unlock();
+      check_tags(p.tag, tag);
       complete_pair(p.buf, buf, p.req_id, id);
     } else {
-      recvs[key].push_back({id, buf});
+      recvs[(int)ch].push_back({id, buf, tag});
     }
     return id;
   }
@@ -294,6 +319,59 @@ c10::intrusive_ptr<c10d::Backend> gloo_backend(const std::string& group_name) {
   TORCH_CHECK(pg, "mpi4torch_amd: process group '", group_name,
               "' not found in c10d registry");
   return pg->getBackend(c10::DeviceType::CPU);
+}
+
+// Host-channel gloo tag layout (disjoint from p2p data tags):
+//   0            exchange() collective-internal block traffic
+//   1 + ch       C10dTransport p2p data, FIFO per (peer, channel)
+//   16 + ch      MPI4TORCH_AMD_DEBUG p2p metadata handshake
+int chan_tag(Channel ch) { return 1 + (int)ch; }
+int meta_tag(Channel ch) { return 16 + (int)ch; }
+
+// MPI4TORCH_AMD_DEBUG=1 p2p handshake: piggyback [tag, dtype, numel] over
+// the host (gloo) channel alongside every p2p transfer. The meta stream is
+// FIFO per (peer, channel) exactly like the data stream, so if user posts
+// cross (e.g. tags (a,b) sent against recvs (b,a)), the receiver's Wait
+// raises with both ranks' views instead of silently delivering swapped
+// payloads. Completion is checked at wait_request — a sync point anyway —
+// so debug mode adds no new blocking before that.
+void attach_debug_handshake(Request& r,
+                            const c10::intrusive_ptr<c10d::Backend>& backend,
+                            bool is_send, const at::Tensor& buf, int peer,
+                            int tag, Channel ch) {
+  if (!config().debug_collectives || !backend) return;
+  auto meta = at::zeros({3}, at::TensorOptions().dtype(at::kLong));
+  std::vector<at::Tensor> ts{meta};
+  if (is_send) {
+    auto* m = meta.data_ptr<int64_t>();
+    m[0] = tag;
+    m[1] = (int64_t)buf.scalar_type();
+    m[2] = buf.numel();
+    r.debug_work = backend->send(ts, peer, meta_tag(ch));
+  } else {
+    r.debug_expect = {(int64_t)tag, (int64_t)buf.scalar_type(), buf.numel()};
+    r.debug_work = backend->recv(ts, peer, meta_tag(ch));
+  }
+  r.debug_meta = meta;
+  r.debug_peer = peer;
+}
+
+void check_debug_handshake(Request& r) {
+  if (!r.debug_work) return;
+  wait_work(r.debug_work);
+  if (r.debug_expect.empty()) return;  // send side: delivery only
+  const auto* m = r.debug_meta.data_ptr<int64_t>();
+  TORCH_CHECK(
+      m[0] == r.debug_expect[0] && m[1] == r.debug_expect[1] &&
+          m[2] == r.debug_expect[2],
+      "mpi4torch_amd[debug]: p2p transfer mismatch with rank ", r.debug_peer,
+      " — matching is FIFO per (peer, channel) and the pair that matched "
+      "disagrees: sender posted (tag=", m[0], ", dtype=",
+      (at::ScalarType)m[1], ", numel=", m[2], ") but this rank's recv "
+      "expected (tag=", r.debug_expect[0], ", dtype=",
+      (at::ScalarType)r.debug_expect[1], ", numel=", r.debug_expect[2],
+      "). Tags do not reorder matching; make both ranks post their "
+      "transfers per peer+channel in the same order.");
 }
 
 // ---------------------------------------------------------------------------
@@ -453,6 +531,10 @@ struct C10dTransport final : Transport {
     }
     for (auto& w : works) wait_work(w);
   }
+  // p2p matching is FIFO per (peer, channel) — the fixed per-channel gloo
+  // tag reproduces the RCCL transport's tag-free semantics exactly, so the
+  // CPU SPMD suite exercises the same contract the GPU runs (user tags are
+  // metadata, validated by the debug handshake).
   uint64_t isend(const at::Tensor& buf, int peer, int tag,
                  Channel ch) override {
     if (peer == rank()) return self_matcher().isend(buf, tag, ch);
@@ -460,7 +542,8 @@ struct C10dTransport final : Transport {
     std::vector<at::Tensor> ts{const_cast<at::Tensor&>(buf)};
     Request r;
     r.buffer = buf;
-    r.work = backend_->send(ts, peer, user_tag(tag, ch));
+    attach_debug_handshake(r, backend_, /*is_send=*/true, buf, peer, tag, ch);
+    r.work = backend_->send(ts, peer, chan_tag(ch));
     return requests().add(std::move(r));
   }
   uint64_t irecv(at::Tensor& buf, int peer, int tag, Channel ch) override {
@@ -469,7 +552,8 @@ struct C10dTransport final : Transport {
     std::vector<at::Tensor> ts{buf};
     Request r;
     r.buffer = buf;
-    r.work = backend_->recv(ts, peer, user_tag(tag, ch));
+    attach_debug_handshake(r, backend_, /*is_send=*/false, buf, peer, tag, ch);
+    r.work = backend_->recv(ts, peer, chan_tag(ch));
     return requests().add(std::move(r));
   }
   uint64_t iallreduce(const at::Tensor& in, at::Tensor& out,
@@ -495,13 +579,6 @@ struct C10dTransport final : Transport {
   }
 
  private:
-  // User p2p tags are offset per channel so forward and adjoint transfers
-  // never match each other (role of the reference's tag+10,
-  // csrc/extension.cpp:1161), and offset by 1 so user tag 0 does not
-  // collide with collective-internal block traffic.
-  static int user_tag(int tag, Channel ch) {
-    return 1 + tag * 4 + (int)ch;
-  }
   std::string name_;
   c10::intrusive_ptr<c10d::Backend> backend_;
   std::mutex mu_;
@@ -521,6 +598,7 @@ struct RcclTransport final : Transport {
                  c10::hip::getStreamFromPoolMasqueradingAsCUDA(true, device),
                  c10::hip::getStreamFromPoolMasqueradingAsCUDA(true, device)} {
     auto backend = gloo_backend(group_name);
+    host_backend_ = backend;  // kept for the debug p2p handshake
     rank_ = backend->getRank();
     size_ = backend->getSize();
     // ncclUniqueId exchange over the gloo backend: the MI355X equivalent of
@@ -641,24 +719,32 @@ struct RcclTransport final : Transport {
           nbytes(*self_sends[i]), hipMemcpyDeviceToDevice, hop.stream()));
     }
   }
+  // Non-blocking p2p is DEFERRED: the nccl call is not issued here. Every
+  // pending send/recv of this transport is launched by flush_p2p() inside
+  // ONE ncclGroupStart/End (triggered by the first Wait on any of them) so
+  // matched pairs land in a single fused kernel and can rendezvous — the
+  // c10d batch_isend_irecv pattern. Eagerly issuing ncclSend/ncclRecv
+  // serially on one stream deadlocks a ring/bidirectional exchange on >=2
+  // GPUs once payloads exceed RCCL's internal FIFO buffering: every rank's
+  // send kernel spins for a peer recv that is queued *behind* a blocked
+  // send on that peer's own p2p stream.
   uint64_t isend(const at::Tensor& buf, int peer, int tag,
                  Channel ch) override {
     if (peer == rank_) return self_matcher().isend(buf, tag, ch);
     std::lock_guard<std::mutex> g(mu_);
     check_not_capturing("Isend");
-    enter_side(ch, {buf});
-    M4A_NCCL_CHECK(ncclSend(buf.data_ptr(), nbytes(buf), ncclUint8, peer,
-                            comm(ch), stream(ch)));
-    return make_gpu_request(ch, buf);
+    return defer_p2p(/*is_send=*/true, buf, peer, tag, ch);
   }
   uint64_t irecv(at::Tensor& buf, int peer, int tag, Channel ch) override {
     if (peer == rank_) return self_matcher().irecv(buf, tag, ch);
     std::lock_guard<std::mutex> g(mu_);
     check_not_capturing("Irecv");
-    enter_side(ch, {buf});
-    M4A_NCCL_CHECK(ncclRecv(buf.data_ptr(), nbytes(buf), ncclUint8, peer,
-                            comm(ch), stream(ch)));
-    return make_gpu_request(ch, buf);
+    return defer_p2p(/*is_send=*/false, buf, peer, tag, ch);
+  }
+
+  void flush_p2p() override {
+    std::lock_guard<std::mutex> g(mu_);
+    flush_pending_locked();
   }
   uint64_t ireduce_scatter(const at::Tensor& in, at::Tensor& out,
                            RedOp op) override {
@@ -784,6 +870,73 @@ struct RcclTransport final : Transport {
     return requests().add(std::move(r));
   }
 
+  struct PendingP2P {
+    bool is_send;
+    at::Tensor buf;
+    int peer;
+    Channel ch;
+    uint64_t req_id;
+    hipEvent_t ready;  // caller-stream position at enqueue time
+  };
+
+  // mu_ held. Record where the caller's stream is (the buffer's producing
+  // ops), park the op, and hand back a deferred request.
+  uint64_t defer_p2p(bool is_send, const at::Tensor& buf, int peer, int tag,
+                     Channel ch) {
+    check_async_errors();
+    auto cur = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(device_);
+    auto& pool = EventPool::forDevice(device_);
+    hipEvent_t ready = pool.acquire();
+    M4A_HIP_CHECK(hipEventRecord(ready, cur.stream()));
+    c10::hip::HIPCachingAllocatorMasqueradingAsCUDA::
+        recordStreamMasqueradingAsCUDA(buf.storage().data_ptr(),
+                                       streams_[(int)ch]);
+    Request r;
+    r.gpu = true;
+    r.device = device_;
+    r.deferred = true;
+    r.buffer = buf;
+    r.owner = weak_from_this();
+    attach_debug_handshake(r, host_backend_, is_send, buf, peer, tag, ch);
+    uint64_t id = requests().add(std::move(r));
+    pending_.push_back({is_send, buf, peer, ch, id, ready});
+    return id;
+  }
+
+  // mu_ held. Launch every pending p2p op in one grouped call (one fused
+  // RCCL kernel per channel), then stamp each request's completion event.
+  void flush_pending_locked() {
+    if (pending_.empty()) return;
+    check_async_errors();
+    auto& pool = EventPool::forDevice(device_);
+    for (auto& p : pending_) {
+      M4A_HIP_CHECK(hipStreamWaitEvent(stream(p.ch), p.ready, 0));
+      pool.release(p.ready);
+    }
+    M4A_NCCL_CHECK(ncclGroupStart());
+    for (auto& p : pending_) {
+      if (nbytes(p.buf) == 0) continue;
+      if (p.is_send) {
+        M4A_NCCL_CHECK(ncclSend(p.buf.data_ptr(), nbytes(p.buf), ncclUint8,
+                                p.peer, comm(p.ch), stream(p.ch)));
+      } else {
+        M4A_NCCL_CHECK(ncclRecv(p.buf.data_ptr(), nbytes(p.buf), ncclUint8,
+                                p.peer, comm(p.ch), stream(p.ch)));
+      }
+    }
+    M4A_NCCL_CHECK(ncclGroupEnd());
+    auto& tab = requests();
+    std::lock_guard<std::mutex> tg(tab.mu);
+    for (auto& p : pending_) {
+      Request* r = tab.peek(p.req_id);
+      if (!r) continue;
+      r->deferred = false;
+      r->event = pool.acquire();
+      M4A_HIP_CHECK(hipEventRecord(r->event, stream(p.ch)));
+    }
+    pending_.clear();
+  }
+
   // Stream bracket for one collective. Under hipGraph capture the op runs
   // directly on the capturing stream (events/recordStream are skipped: the
   // graph serializes ordering and graph memory pools own lifetimes) so
@@ -816,15 +969,33 @@ struct RcclTransport final : Transport {
   ncclComm_t comms_[3] = {nullptr, nullptr, nullptr};
   int fp8_state_[2] = {0, 0};  // 0 unknown, 1 native, -1 cast fallback
   c10::hip::HIPStreamMasqueradingAsCUDA streams_[3];
+  c10::intrusive_ptr<c10d::Backend> host_backend_;
+  std::vector<PendingP2P> pending_;
 };
 
 } // namespace
 
 void wait_request(uint64_t id) {
+  // Deferred RCCL p2p: launch the owning transport's whole pending batch
+  // (one grouped call) before completing this request.
+  {
+    std::shared_ptr<Transport> owner;
+    {
+      auto& tab = requests();
+      std::lock_guard<std::mutex> g(tab.mu);
+      Request* r = tab.peek(id);
+      TORCH_CHECK(r, "mpi4torch_amd: unknown or already-waited request id ",
+                  id);
+      if (r->deferred) owner = r->owner.lock();
+    }
+    if (owner) owner->flush_p2p();
+  }
   Request r = requests().take(id);
   TORCH_CHECK(!r.self_pending,
               "mpi4torch_amd: Wait() on a self send/recv whose matching "
               "operation was never posted");
+  TORCH_CHECK(!r.deferred,
+              "mpi4torch_amd internal: deferred p2p request not flushed");
   if (r.gpu) {
     auto cur = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(r.device);
     M4A_HIP_CHECK(hipStreamWaitEvent(cur.stream(), r.event, 0));
@@ -832,6 +1003,7 @@ void wait_request(uint64_t id) {
   } else if (r.work) {
     wait_work(r.work);
   }
+  check_debug_handshake(r);
 }
 
 std::shared_ptr<Transport> make_local_transport() {

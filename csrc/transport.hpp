@@ -34,11 +34,20 @@ namespace m4a {
 
 enum class Channel : int { Coll = 0, P2P = 1, P2PBwd = 2 };
 
-struct Transport {
+struct Transport : std::enable_shared_from_this<Transport> {
   virtual ~Transport() = default;
   virtual int rank() const = 0;
   virtual int size() const = 0;
   virtual bool is_gpu() const = 0;
+
+  // Launch any deferred point-to-point operations (see RcclTransport: p2p
+  // enqueues are batched and issued as ONE ncclGroupStart/End at the first
+  // Wait, c10d batch_isend_irecv style, so matched send/recv pairs can
+  // rendezvous — serially-issued ncclSend/ncclRecv on one stream deadlock
+  // in a ring once payloads exceed RCCL's internal buffering). No-op on
+  // transports whose p2p is eagerly posted (gloo has a host-side matching
+  // engine, like MPI's).
+  virtual void flush_p2p() {}
 
   // All tensors must be contiguous and on this transport's device class.
   // GPU: stream-ordered with respect to the caller's current stream
@@ -64,6 +73,12 @@ struct Transport {
                         std::vector<at::Tensor>& recvbufs,
                         const std::vector<int>& rpeers) = 0;
   // Non-blocking p2p. Returns a request id resolvable via wait_request().
+  // Matching contract (identical on ALL transports): FIFO per
+  // (peer, channel) — the n-th send posted to a peer on a channel matches
+  // the n-th recv posted from it on that channel. `tag` does NOT
+  // disambiguate matching (RCCL has no tags); it is metadata, validated
+  // against the peer's tag under MPI4TORCH_AMD_DEBUG=1 so crossed
+  // transfers raise instead of silently swapping payloads.
   virtual uint64_t isend(const at::Tensor& buf, int peer, int tag,
                          Channel ch) = 0;
   virtual uint64_t irecv(at::Tensor& buf, int peer, int tag, Channel ch) = 0;

@@ -996,6 +996,10 @@ void wait_request(uint64_t id) {
               "operation was never posted");
   TORCH_CHECK(!r.deferred,
               "mpi4torch_amd internal: deferred p2p request not flushed");
+  // Recv side validates the handshake BEFORE the data wait: the sender's
+  // metadata arrives no later than its payload, so a mismatched pair is
+  // diagnosed even when the data transfer itself would hang.
+  if (!r.debug_expect.empty()) check_debug_handshake(r);
   if (r.gpu) {
     auto cur = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(r.device);
     M4A_HIP_CHECK(hipStreamWaitEvent(cur.stream(), r.event, 0));
@@ -1003,7 +1007,9 @@ void wait_request(uint64_t id) {
   } else if (r.work) {
     wait_work(r.work);
   }
-  check_debug_handshake(r);
+  // Send side checks after the data wait (data delivered => the peer posted
+  // its recv => its meta recv is posted too; no added blocking).
+  if (r.debug_expect.empty()) check_debug_handshake(r);
 }
 
 std::shared_ptr<Transport> make_local_transport() {

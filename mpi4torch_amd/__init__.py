@@ -243,12 +243,24 @@ class MPI_Communicator:
         return WaitHandle(self._comm.Ireducescatter(tensor, op))
 
     def Isend(self, tensor: torch.Tensor, dest: int, tag: int) -> WaitHandle:
-        """Non-blocking send; complete with Wait. Backward: reverse recv."""
+        """Non-blocking send; complete with Wait. Backward: reverse recv.
+
+        Matching contract (ALL transports): FIFO per (peer, channel) — the
+        n-th send posted to a peer matches its n-th recv from this rank.
+        Unlike MPI, ``tag`` does NOT reorder matching (RCCL has no tags);
+        it is metadata, validated against the receiver's tag under
+        MPI4TORCH_AMD_DEBUG=1 so crossed transfers raise instead of
+        silently swapping payloads.
+        """
         return WaitHandle(self._comm.Isend(tensor, dest, tag))
 
     def Irecv(self, tensor: torch.Tensor, source: int, tag: int) -> WaitHandle:
         """Non-blocking receive into ``tensor``'s buffer; complete with
-        Wait. Backward: reverse send."""
+        Wait. Backward: reverse send.
+
+        Matching is FIFO per (peer, channel); ``tag`` is validating
+        metadata only — see Isend.
+        """
         return WaitHandle(self._comm.Irecv(tensor, source, tag))
 
     def Wait(self, waithandle: WaitHandle) -> torch.Tensor:
@@ -336,8 +348,8 @@ def comm_from_process_group(pg) -> MPI_Communicator:
     The torch.distributed analog of the reference's comm_from_mpi4py
     (src/__init__.py:247-261): sub-communicators are torch.distributed
     groups. If the group's backend has no CPU (gloo) path, a companion gloo
-    group over the same ranks is created (collective call: all ranks of the
-    default world must reach this point).
+    group over the same ranks is created (collective over the group's
+    members only — every member must reach this point).
     """
     import torch.distributed as dist
 
@@ -346,7 +358,8 @@ def comm_from_process_group(pg) -> MPI_Communicator:
         name = pg.group_name
     except Exception:
         ranks = dist.get_process_group_ranks(pg)
-        companion = dist.new_group(ranks=ranks, backend="gloo")
+        companion = dist.new_group(ranks=ranks, backend="gloo",
+                                   use_local_synchronization=True)
         name = companion.group_name
     return MPI_Communicator(torch.ops.mpi4torch_amd.comm_from_group(name))
 
@@ -366,19 +379,26 @@ def comm_split(comm: MPI_Communicator, color: int) -> Optional[MPI_Communicator]
     import torch.distributed as dist
 
     world = comm.size
-    colors_t = comm.Allgather(
-        torch.tensor([float(color)], dtype=torch.float64), 0)
-    colors = [int(colors_t[r].item()) for r in range(world)]
-    mine = None
-    # every rank must create EVERY subgroup (new_group is collective over
-    # the default world)
-    for c in sorted(set(c for c in colors if c >= 0)):
-        ranks = [r for r in range(world) if colors[r] == c]
-        g = dist.new_group(ranks=ranks, backend="gloo")
-        if c == color:
-            mine = g
-    if color < 0 or mine is None:
+    if world == 1 or not (dist.is_available() and dist.is_initialized()):
+        # splitting a singleton / local communicator is the identity
+        return comm if color >= 0 else None
+    # dist.new_group takes GLOBAL ranks (of the default world), while this
+    # comm's members may already be a subgroup with renumbered local ranks —
+    # so allgather (color, global_rank) pairs over `comm` and pass the
+    # global ranks through. use_local_synchronization makes new_group
+    # collective over the subgroup only, so splitting a split communicator
+    # works (non-members never call it).
+    me_global = dist.get_rank()
+    pairs_t = comm.Allgather(
+        torch.tensor([float(color), float(me_global)], dtype=torch.float64), 0)
+    pairs = pairs_t.view(world, 2)
+    colors = [int(pairs[r, 0].item()) for r in range(world)]
+    granks = [int(pairs[r, 1].item()) for r in range(world)]
+    if color < 0:
         return None
+    ranks = sorted(granks[r] for r in range(world) if colors[r] == color)
+    mine = dist.new_group(ranks=ranks, backend="gloo",
+                          use_local_synchronization=True)
     return comm_from_process_group(mine)
 
 

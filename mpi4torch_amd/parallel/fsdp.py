@@ -125,7 +125,11 @@ class FullyShardedDataParallel(torch.nn.Module):
         if self.comm.size > 1:
             with torch.no_grad():
                 for p in module.parameters():
-                    self.comm.Bcast_(p.data, 0)
+                    # Bcast_ returns a fresh tensor when an internal copy
+                    # was needed (non-contiguous / host staging): copy back
+                    res = self.comm.Bcast_(p.data, 0)
+                    if res.data_ptr() != p.data.data_ptr():
+                        p.data.copy_(res)
         unit_modules = list(units) if units is not None else [
             m for m in module.children()
             if any(p.requires_grad for p in m.parameters())

@@ -72,7 +72,10 @@ class GPipe:
 
         outputs: List[torch.Tensor] = []
         losses: List[torch.Tensor] = []
-        # ---- forward fill: microbatch i rides tag i on the p2p channel
+        # ---- forward fill: microbatch i is labeled tag i (metadata only —
+        # matching is FIFO per (peer, channel); the fill loop posts
+        # microbatches in the same order on every rank, which is what
+        # actually keeps them paired. MPI4TORCH_AMD_DEBUG=1 validates.)
         for i in range(n):
             if self.is_first:
                 x = microbatches[i].to(device)

@@ -46,7 +46,11 @@ class ShardedDataParallel(torch.nn.Module):
         if P > 1:
             with torch.no_grad():
                 for p in self.module.parameters():
-                    self.comm.Bcast_(p.data, 0)
+                    # Bcast_ returns a fresh tensor when an internal copy
+                    # was needed (non-contiguous / host staging): copy back
+                    res = self.comm.Bcast_(p.data, 0)
+                    if res.data_ptr() != p.data.data_ptr():
+                        p.data.copy_(res)
 
         params = [p for p in self.module.parameters() if p.requires_grad]
         assert params, "no trainable parameters"

@@ -42,6 +42,80 @@ def test_debug_off_ws2():
     run_spmd(2, _debug_off_worker)
 
 
+def _p2p_tag_cross_worker(rank, world):
+    os.environ["MPI4TORCH_AMD_DEBUG"] = "1"
+    import mpi4torch_amd as m
+
+    comm = m.COMM_WORLD
+    # rank 0 posts tags (1, 2); rank 1 recvs in order (2, 1). Matching is
+    # FIFO per (peer, channel), so the pairs cross — the debug handshake
+    # must turn the silent payload swap into an error at Wait.
+    if rank == 0:
+        h1 = comm.Isend(torch.ones(8), 1, 1)
+        h2 = comm.Isend(torch.ones(8) * 2, 1, 2)
+        comm.Wait(h1)
+        comm.Wait(h2)
+    else:
+        r1 = comm.Irecv(torch.empty(8), 0, 2)
+        r2 = comm.Irecv(torch.empty(8), 0, 1)
+        try:
+            comm.Wait(r1)
+            raise AssertionError("expected p2p tag-mismatch detection")
+        except RuntimeError as e:
+            assert "FIFO" in str(e) and "tag=1" in str(e), e
+        try:
+            comm.Wait(r2)  # the second pair is crossed too
+            raise AssertionError("expected p2p tag-mismatch detection")
+        except RuntimeError as e:
+            assert "tag=2" in str(e), e
+
+
+def test_p2p_tag_cross_ws2():
+    run_spmd(2, _p2p_tag_cross_worker)
+
+
+def _p2p_dtype_mismatch_worker(rank, world):
+    os.environ["MPI4TORCH_AMD_DEBUG"] = "1"
+    import mpi4torch_amd as m
+
+    comm = m.COMM_WORLD
+    # same byte count, different dtype: the wire moves bytes happily; only
+    # the handshake can catch it
+    if rank == 0:
+        h = comm.Isend(torch.ones(8, dtype=torch.float32), 1, 0)
+        comm.Wait(h)
+    else:
+        r = comm.Irecv(torch.empty(8, dtype=torch.int32), 0, 0)
+        try:
+            comm.Wait(r)
+            raise AssertionError("expected p2p dtype-mismatch detection")
+        except RuntimeError as e:
+            assert "dtype" in str(e), e
+
+
+def test_p2p_dtype_mismatch_ws2():
+    run_spmd(2, _p2p_dtype_mismatch_worker)
+
+
+def _p2p_matched_with_debug_worker(rank, world):
+    os.environ["MPI4TORCH_AMD_DEBUG"] = "1"
+    import mpi4torch_amd as m
+
+    comm = m.COMM_WORLD
+    # a correct ring still works with the handshake on
+    src = torch.full((64,), float(rank))
+    h = comm.Isend(src, (rank + 1) % world, 7)
+    got = comm.Recv(
+        m.JoinDummies(torch.empty(64), [h.dummy]), (rank + world - 1) % world, 7
+    )
+    comm.Wait(m.JoinDummiesHandle(h, [got]))
+    assert (got == (rank + world - 1) % world).all()
+
+
+def test_p2p_matched_with_debug_ws5():
+    run_spmd(5, _p2p_matched_with_debug_worker)
+
+
 def _timeout_worker(rank, world):
     import mpi4torch_amd as m
 

@@ -109,3 +109,34 @@ def _split_worker(rank, world):
 
 def test_comm_split_ws5():
     run_spmd(5, _split_worker)
+
+
+def _grandchild_split_worker(rank, world):
+    import mpi4torch_amd as m
+
+    comm = m.COMM_WORLD
+    # first split: {0,1,2,3} vs {4,5,6}
+    c1 = 0 if rank < 4 else 1
+    child = m.comm_split(comm, c1)
+    members1 = [r for r in range(world) if (0 if r < 4 else 1) == c1]
+    assert child.size == len(members1)
+    assert child.rank == members1.index(rank)
+
+    # split the CHILD: global-rank parity. Round 1's bug passed child-local
+    # ranks to dist.new_group (which wants GLOBAL ranks), silently grouping
+    # the wrong processes for any comm that is not the world.
+    c2 = rank % 2
+    grand = m.comm_split(child, c2)
+    members2 = [r for r in members1 if r % 2 == c2]
+    assert grand.size == len(members2), (rank, grand.size, members2)
+    assert grand.rank == members2.index(rank), (rank, grand.rank)
+    t = torch.full((3,), float(rank), dtype=torch.double)
+    s = grand.Allreduce(t, m.MPI_SUM)
+    assert (s == sum(members2)).all(), (rank, s[0].item(), members2)
+    # and the grandchild is a working communicator for axis collectives too
+    g = grand.Allgather(torch.full((1,), float(rank), dtype=torch.double), 0)
+    assert [int(v.item()) for v in g] == members2
+
+
+def test_comm_split_grandchild_ws7():
+    run_spmd(7, _grandchild_split_worker)

@@ -509,3 +509,76 @@ def test_zero2_accumulation_ws2():
 
 def test_fsdp_accumulation_ws2():
     run_spmd(2, _fsdp_accum_worker)
+
+
+class _BranchyNet(torch.nn.Module):
+    """Shared trunk + a branch only some ranks execute."""
+
+    def __init__(self):
+        super().__init__()
+        self.trunk = torch.nn.Linear(8, 8, bias=False)
+        self.extra = torch.nn.Linear(8, 8, bias=False)
+
+    def forward(self, x, use_extra: bool):
+        y = self.trunk(x)
+        if use_extra:
+            y = y + self.extra(x)
+        return y
+
+
+def _ddp_unused_raises_worker(rank, world):
+    from mpi4torch_amd.parallel import DistributedDataParallel
+
+    torch.manual_seed(3)
+    # cap 0 => one bucket per parameter, so the trunk's bucket fires while
+    # extra's never fills
+    model = DistributedDataParallel(_BranchyNet(), bucket_cap_mb=0)
+    x = torch.randn(4, 8)
+    # every rank skips `extra`: its bucket never fills while the trunk's
+    # bucket launched — default mode must raise loudly, not silently skip
+    model(x, use_extra=False).sum().backward()
+    try:
+        model.finish_gradient_sync()
+        raise AssertionError("expected unused-parameter error")
+    except RuntimeError as e:
+        assert "find_unused_parameters" in str(e), e
+
+
+def test_ddp_unused_raises_ws3():
+    run_spmd(3, _ddp_unused_raises_worker)
+
+
+def _ddp_find_unused_worker(rank, world):
+    import mpi4torch_amd as m
+    from mpi4torch_amd.parallel import DistributedDataParallel
+
+    torch.manual_seed(3)
+    net = _BranchyNet()
+    model = DistributedDataParallel(net, bucket_cap_mb=1,
+                                    find_unused_parameters=True)
+    torch.manual_seed(100 + rank)
+    x = torch.randn(4, 8)
+    use_extra = rank % 2 == 0  # conditionally-used branch, rank-dependent
+    model(x, use_extra).sum().backward()
+    local_trunk = net.trunk.weight.grad.clone()
+    local_extra = (net.extra.weight.grad.clone()
+                   if use_extra else torch.zeros(8, 8))
+    model.finish_gradient_sync()
+
+    comm = m.COMM_WORLD
+    want_trunk = comm.Allreduce(local_trunk, m.MPI_SUM) / world
+    want_extra = comm.Allreduce(local_extra, m.MPI_SUM) / world
+    assert torch.allclose(net.trunk.weight.grad, want_trunk, atol=1e-6)
+    # ranks that skipped the branch get the globally-averaged gradient too
+    assert net.extra.weight.grad is not None
+    assert torch.allclose(net.extra.weight.grad, want_extra, atol=1e-6)
+
+    # a second iteration with the roles swapped still lines up
+    for p in net.parameters():
+        p.grad = None
+    model(x, not use_extra).sum().backward()
+    model.finish_gradient_sync()
+
+
+def test_ddp_find_unused_ws3():
+    run_spmd(3, _ddp_find_unused_worker)

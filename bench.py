@@ -28,6 +28,13 @@ import torch
 
 
 def setup_world(args):
+    # Time REAL communication even at N=1: force_full_path disables the
+    # world-1 local fast paths so the timed region contains an actual
+    # ncclAllReduce / grouped RCCL exchange (cost parity with the local
+    # slab-kernel path was measured in profiles/microbench_final.txt;
+    # rocprof on this bench shows the RCCL kernels). Set before the
+    # extension reads its config.
+    os.environ.setdefault("MPI4TORCH_AMD_FORCE_FULL_PATH", "1")
     if "WORLD_SIZE" not in os.environ:
         os.environ["MASTER_ADDR"] = "127.0.0.1"
         os.environ["MASTER_PORT"] = os.environ.get("MASTER_PORT", "29400")
@@ -219,12 +226,23 @@ def main():
     p.add_argument("--config", default="allreduce", choices=sorted(BENCHES))
     p.add_argument("--size-mib", type=float, default=1024.0,
                    help="allreduce tensor size per GPU (MiB)")
+    p.add_argument("--all-configs", action="store_true",
+                   help="run every BASELINE config; write all JSON records "
+                        "to gpurun_out/bench_all.jsonl, print the headline "
+                        "(allreduce) line last")
     args = p.parse_args()
 
     m4a, rank, world, device = setup_world(args)
     comm = m4a.COMM_WORLD
     assert comm.size == world
 
+    extra_records = []
+    if args.all_configs:
+        for name in ("linreg", "ring", "alltoall"):
+            r, ts = BENCHES[name](m4a, comm, device, args)
+            r["ms_per_step"] = round(ts * 1e3, 3)
+            extra_records.append(r)
+        args.config = "allreduce"
     result, t_step = BENCHES[args.config](m4a, comm, device, args)
     # RCCL's version banner is written through C stdio, which is fully
     # buffered on a pipe and flushes at exit — AFTER our JSON. Flush it now
@@ -235,6 +253,11 @@ def main():
         ctypes.CDLL(None).fflush(None)
     except Exception:
         pass
+    if rank == 0 and extra_records:
+        os.makedirs("gpurun_out", exist_ok=True)
+        with open("gpurun_out/bench_all.jsonl", "w") as f:
+            for r in extra_records:
+                f.write(json.dumps({**r, "n_gpus": world}) + "\n")
     if rank == 0:
         out = {
             "metric": result["metric"],

@@ -506,4 +506,90 @@ void launch_fp8_reduce(const void* in, void* out, int64_t n_elems, int nranks,
   }
 }
 
+// ---------------------------------------------------------------------------
+// MINLOC/MAXLOC pair arg-reduction. RCCL (like NCCL) has no MPI pair types;
+// the op layer allgathers the (value, location) pairs and this kernel does
+// the arg-reduce locally — completing the reference's 12-op table
+// (reference csrc/extension.cpp:204-252). Tie-break: smallest location
+// (MPI-defined).
+// ---------------------------------------------------------------------------
+
+#include <hip/hip_fp16.h>
+#include <hip/hip_bfloat16.h>
+
+namespace {
+
+template <typename T>
+__device__ inline bool pl_lt(T a, T b) { return a < b; }
+template <>
+__device__ inline bool pl_lt<__half>(__half a, __half b) {
+  return (float)a < (float)b;
+}
+template <>
+__device__ inline bool pl_lt<hip_bfloat16>(hip_bfloat16 a, hip_bfloat16 b) {
+  return (float)a < (float)b;
+}
+template <typename T>
+__device__ inline bool pl_eq(T a, T b) { return !pl_lt(a, b) && !pl_lt(b, a); }
+
+template <typename T, bool MAXLOC>
+__global__ __launch_bounds__(256) void pairloc_reduce_kernel(
+    const T* __restrict__ in, T* __restrict__ out, long long n,
+    long long nranks) {
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n; i += stride) {
+    T bv = in[2 * i];
+    T bl = in[2 * i + 1];
+    for (long long r = 1; r < nranks; ++r) {
+      const T v = in[r * 2 * n + 2 * i];
+      const T l = in[r * 2 * n + 2 * i + 1];
+      const bool better = MAXLOC ? pl_lt(bv, v) : pl_lt(v, bv);
+      if (better || (pl_eq(v, bv) && pl_lt(l, bl))) {
+        bv = v;
+        bl = l;
+      }
+    }
+    out[2 * i] = bv;
+    out[2 * i + 1] = bl;
+  }
+}
+
+template <typename T>
+void launch_pairloc_t(const void* in, void* out, int64_t n, int nranks,
+                      int op, hipStream_t stream) {
+  long long blocks = (n + 255) / 256;
+  if (blocks > 4096) blocks = 4096;
+  if (blocks < 1) blocks = 1;
+  if (op == 0) {
+    hipLaunchKernelGGL((pairloc_reduce_kernel<T, false>),
+                       dim3((unsigned)blocks), dim3(256), 0, stream,
+                       static_cast<const T*>(in), static_cast<T*>(out),
+                       (long long)n, (long long)nranks);
+  } else {
+    hipLaunchKernelGGL((pairloc_reduce_kernel<T, true>),
+                       dim3((unsigned)blocks), dim3(256), 0, stream,
+                       static_cast<const T*>(in), static_cast<T*>(out),
+                       (long long)n, (long long)nranks);
+  }
+}
+
+} // namespace
+
+void launch_pairloc_reduce(const void* in, void* out, int64_t n_pairs,
+                           int nranks, int op, int dtype,
+                           hipStream_t stream) {
+  switch (dtype) {
+    case 0: launch_pairloc_t<float>(in, out, n_pairs, nranks, op, stream); break;
+    case 1: launch_pairloc_t<double>(in, out, n_pairs, nranks, op, stream); break;
+    case 2: launch_pairloc_t<__half>(in, out, n_pairs, nranks, op, stream); break;
+    case 3: launch_pairloc_t<hip_bfloat16>(in, out, n_pairs, nranks, op, stream); break;
+    case 4: launch_pairloc_t<signed char>(in, out, n_pairs, nranks, op, stream); break;
+    case 5: launch_pairloc_t<unsigned char>(in, out, n_pairs, nranks, op, stream); break;
+    case 6: launch_pairloc_t<short>(in, out, n_pairs, nranks, op, stream); break;
+    case 7: launch_pairloc_t<int>(in, out, n_pairs, nranks, op, stream); break;
+    default: launch_pairloc_t<long long>(in, out, n_pairs, nranks, op, stream); break;
+  }
+}
+
 } // namespace m4a

@@ -46,6 +46,16 @@ void launch_slab_copy(const SlabDesc* descs, int n, hipStream_t stream);
 void launch_bitwise_reduce(const void* in, void* out, int64_t chunk_bytes,
                            int nranks, int op, hipStream_t stream);
 
+// MINLOC/MAXLOC local arg-reduction across `nranks` contiguous chunks of
+// (value, location) pairs (MPI pair-type semantics, reference op table
+// csrc/extension.cpp:204-252): in is [nranks][n_pairs][2] (last axis =
+// value, location in the same dtype), out is [n_pairs][2].
+//   op 0 = MINLOC: smallest value, ties -> smallest location
+//   op 1 = MAXLOC: largest value, ties -> smallest location
+// dtype codes: 0=f32 1=f64 2=f16 3=bf16 4=i8 5=u8 6=i16 7=i32 8=i64
+void launch_pairloc_reduce(const void* in, void* out, int64_t n_pairs,
+                           int nranks, int op, int dtype, hipStream_t stream);
+
 // Fused fp8 local reduction with fp32 accumulation: elementwise
 //   out[i] = op_{r<nranks} fp32(in[r*n + i])  quantized to fp8 ONCE.
 // This is the MI355X fp8 allreduce tail (allgather + this kernel): one

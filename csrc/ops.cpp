@@ -35,6 +35,7 @@
 #include <algorithm>
 #include <cstdlib>
 #include <cstring>
+#include <limits>
 #include <map>
 #include <mutex>
 #include <tuple>
@@ -227,18 +228,18 @@ bool is_arith(int64_t op) {
 // Collective-desync detector (config().debug_collectives): compare a
 // signature of this collective across all ranks over the host channel and
 // fail with a description instead of deadlocking in RCCL/gloo.
-void debug_check_collective(const std::string& group, const char* opname,
-                            const at::Tensor& t,
-                            std::initializer_list<int64_t> args) {
-  if (!config().debug_collectives || group.empty()) return;
-  size_t h = std::hash<std::string>()(opname);
-  auto mix = [&h](int64_t v) {
+struct SigHash {
+  size_t h;
+  explicit SigHash(const char* opname) : h(std::hash<std::string>()(opname)) {}
+  void mix(int64_t v) {
     h ^= std::hash<int64_t>()(v) + 0x9e3779b97f4a7c15ull + (h << 6) + (h >> 2);
-  };
-  mix((int64_t)t.scalar_type());
-  for (auto s : t.sizes()) mix(s);
-  for (auto a : args) mix(a);
-  auto all = host_allgather_int64(group, (int64_t)(h & 0x7fffffffffffffffll));
+  }
+};
+
+void debug_compare_sig(const std::string& group, const char* opname,
+                       SigHash& sig) {
+  auto all =
+      host_allgather_int64(group, (int64_t)(sig.h & 0x7fffffffffffffffll));
   for (size_t r = 1; r < all.size(); ++r) {
     TORCH_CHECK(all[r] == all[0],
                 "mpi4torch_amd[debug]: collective desync detected at ", opname,
@@ -248,14 +249,57 @@ void debug_check_collective(const std::string& group, const char* opname,
   }
 }
 
+// Full-shape variant: every dimension must agree across ranks
+// (elementwise collectives: Allreduce, Bcast_, Reduce_, Reducescatter).
+void debug_check_collective(const std::string& group, const char* opname,
+                            const at::Tensor& t,
+                            std::initializer_list<int64_t> args) {
+  if (!config().debug_collectives || group.empty()) return;
+  SigHash sig(opname);
+  sig.mix((int64_t)t.scalar_type());
+  for (auto s : t.sizes()) sig.mix(s);
+  for (auto a : args) sig.mix(a);
+  debug_compare_sig(group, opname, sig);
+}
+
+// Axis-collective variant: per-rank sizes along the listed axes
+// legitimately differ (variable counts), so the signature covers dtype,
+// ndim, every OTHER dimension, and the op arguments.
+void debug_check_axis_collective(const std::string& group, const char* opname,
+                                 const at::Tensor& t,
+                                 std::initializer_list<int64_t> axes,
+                                 std::initializer_list<int64_t> args) {
+  if (!config().debug_collectives || group.empty()) return;
+  SigHash sig(opname);
+  sig.mix((int64_t)t.scalar_type());
+  sig.mix(t.dim());
+  for (int64_t i = 0; i < t.dim(); ++i) {
+    bool skip = false;
+    for (auto a : axes) skip = skip || a == i;
+    sig.mix(skip ? -1 : t.size(i));
+  }
+  for (auto a : args) sig.mix(a);
+  debug_compare_sig(group, opname, sig);
+}
+
+// Dtype-only variant: shapes are rank-local by contract (Scatter's
+// non-root inputs are placeholders whose only binding property is the
+// dtype the output is allocated with).
+void debug_check_dtype_collective(const std::string& group,
+                                  const char* opname, at::ScalarType dtype,
+                                  std::initializer_list<int64_t> args) {
+  if (!config().debug_collectives || group.empty()) return;
+  SigHash sig(opname);
+  sig.mix((int64_t)dtype);
+  for (auto a : args) sig.mix(a);
+  debug_compare_sig(group, opname, sig);
+}
+
 void check_op(int64_t op) {
   TORCH_CHECK(op >= kMax && op <= kMaxLoc, "invalid reduction op ", op);
-  TORCH_CHECK(op != kMinLoc && op != kMaxLoc,
-              "mpi4torch_amd: ", red_op_name(op),
-              " is not supported (MPI pair types have no PyTorch tensor "
-              "equivalent; the reference would fail on dtype mapping too, "
-              "cf. mpi4torch csrc/extension.cpp:106-129)");
 }
+
+bool is_pairloc(int64_t op) { return op == kMinLoc || op == kMaxLoc; }
 
 bool native_reduce_dtype(const Transport& tr, at::ScalarType t) {
   if (tr.is_gpu()) {
@@ -320,12 +364,151 @@ bool w1_shortcut(const Transport& tr) {
   return tr.size() == 1 && !config().force_full_path;
 }
 
+std::vector<int> iota_peers_fwd(int n) {
+  std::vector<int> p(n);
+  for (int i = 0; i < n; ++i) p[i] = i;
+  return p;
+}
+
+// Hierarchical GPU allreduce for ops RCCL cannot reduce on the wire
+// (bitwise, fp8, minloc/maxloc): partition the flat tensor into P
+// near-equal contiguous blocks (aligned to `unit` elements), exchange each
+// rank's copy of block j to rank j in ONE grouped p2p launch, locally
+// reduce the P copies of the owned block with the fused CDNA4 kernel, then
+// allgather the reduced blocks (in place). Versus the round-1 design
+// (allgather everything + local reduce) this caps staging at ~numel
+// instead of P*numel and wire bytes at 2*(P-1)/P*numel instead of
+// (P-1)*numel, while keeping the single-reduction property (each element
+// is combined exactly once, fp8 quantized exactly once).
+template <typename LocalReduce>
+Tensor hierarchical_allreduce(Transport& tr, const Tensor& in, int64_t unit,
+                              LocalReduce local_reduce) {
+  const int P = tr.size();
+  const int me = tr.rank();
+  const int64_t N = in.numel();
+  const int64_t nunits = N / unit;
+  std::vector<int64_t> counts(P);
+  for (int p = 0; p < P; ++p) {
+    counts[p] = (nunits / P + (p < nunits % P ? 1 : 0)) * unit;
+  }
+  auto displs = prefix_displs(counts);
+  auto flat = in.view({-1});
+  auto out = at::empty_like(in);
+  auto out_flat = out.view({-1});
+  auto nz_narrow = [](const Tensor& t, int64_t off, int64_t len) {
+    return len > 0 ? t.narrow(0, off, len) : t.narrow(0, 0, 0);
+  };
+  // phase 1: block exchange — my copy of block j goes to rank j
+  auto staging = at::empty({(int64_t)P * counts[me]}, in.options());
+  std::vector<Tensor> sends(P), recvs(P);
+  for (int p = 0; p < P; ++p) {
+    sends[p] = nz_narrow(flat, displs[p], counts[p]);
+    recvs[p] = nz_narrow(staging, (int64_t)p * counts[me], counts[me]);
+  }
+  auto peers = iota_peers_fwd(P);
+  tr.exchange(sends, peers, recvs, peers);
+  // phase 2: fused local reduction of the P copies of the owned block
+  auto myblock = nz_narrow(out_flat, displs[me], counts[me]);
+  if (counts[me] > 0) local_reduce(staging, myblock, counts[me], P);
+  // phase 3: allgather the reduced blocks. Equal counts: one RCCL
+  // allgather, in place (myblock aliases out_flat at the rank-major
+  // offset — the NCCL in-place convention). Unequal: grouped p2p.
+  if (nunits % P == 0) {
+    tr.allgather_equal(myblock, out_flat);
+  } else {
+    std::vector<Tensor> s2, r2;
+    std::vector<int> sp, rp;
+    for (int p = 0; p < P; ++p) {
+      if (p == me) continue;
+      if (counts[me] > 0) {
+        s2.push_back(myblock);
+        sp.push_back(p);
+      }
+      if (counts[p] > 0) {
+        r2.push_back(out_flat.narrow(0, displs[p], counts[p]));
+        rp.push_back(p);
+      }
+    }
+    tr.exchange(s2, sp, r2, rp);
+  }
+  return out;
+}
+
+int pairloc_dtype_code(at::ScalarType t) {
+  switch (t) {
+    case at::kFloat: return 0;
+    case at::kDouble: return 1;
+    case at::kHalf: return 2;
+    case at::kBFloat16: return 3;
+    case at::kChar: return 4;
+    case at::kByte: return 5;
+    case at::kShort: return 6;
+    case at::kInt: return 7;
+    case at::kLong: return 8;
+    default:
+      TORCH_CHECK(false, "mpi4torch_amd: MPI_MINLOC/MAXLOC unsupported for "
+                  "dtype ", t);
+  }
+}
+
+// Torch-composite pair arg-reduce over stk [n_chunks, n_pairs, 2]: the CPU
+// path and the reference semantics the CDNA4 kernel is tested against.
+Tensor pairloc_local_reduce(const Tensor& stk, int64_t kop,
+                            at::ScalarType out_dtype) {
+  auto vals = stk.select(2, 0);
+  auto locs = stk.select(2, 1).to(at::kDouble);
+  Tensor bestv =
+      kop == 0 ? std::get<0>(vals.min(0)) : std::get<0>(vals.max(0));
+  auto mask = vals.eq(bestv.unsqueeze(0));
+  auto masked = locs.masked_fill(mask.logical_not(),
+                                 std::numeric_limits<double>::infinity());
+  auto bestl = std::get<0>(masked.min(0)).to(out_dtype);
+  return at::stack({bestv, bestl}, -1);
+}
+
+// MINLOC/MAXLOC (reference op table csrc/extension.cpp:204-252): tensors
+// are (value, location) pairs along the LAST axis (size 2), the tensor
+// analog of MPI's pair types (MPI_DOUBLE_INT etc.). Result per pair: the
+// extreme value across ranks and its location, ties -> smallest location.
+Tensor pairloc_lowered(Transport& tr, const Tensor& in, int64_t op) {
+  const int kop = op == kMinLoc ? 0 : 1;
+  const int P = tr.size();
+  const int dt = pairloc_dtype_code(in.scalar_type());  // dtype validation
+  if (tr.is_gpu()) {
+    return hierarchical_allreduce(
+        tr, in, /*unit=*/2,
+        [&](const Tensor& stg, Tensor& blk, int64_t n, int nr) {
+          launch_pairloc_reduce(stg.data_ptr(), blk.data_ptr(), n / 2, nr,
+                                kop, dt, current_gpu_stream(in));
+        });
+  }
+  // CPU: allgather + torch composite arg-reduce
+  auto staging = at::empty({(int64_t)P * in.numel()}, in.options());
+  tr.allgather_equal(in, staging);
+  return pairloc_local_reduce(staging.view({P, -1, 2}), kop,
+                              in.scalar_type())
+      .view(in.sizes())
+      .contiguous();
+}
+
 Tensor allreduce_lowered(Transport& tr, const Tensor& in, int64_t op) {
   if (in.numel() == 0) return in.clone();
-  if (tr.size() == 1 &&
-      (!config().force_full_path || !tr.is_gpu() || !is_arith(op) ||
-       !native_reduce_dtype(tr, in.scalar_type()))) {
-    return fast_clone(in);
+  if (tr.size() == 1) {
+    // force_full_path exercises the real machinery even alone: the native
+    // RCCL ring for arithmetic ops, the hierarchical exchange+kernel path
+    // for bitwise/fp8/pairloc (P=1 self-exchange -> kernel -> in-place
+    // allgather). Everything else short-circuits to a device clone.
+    const bool fp8_in = in.scalar_type() == at::kFloat8_e4m3fn ||
+                        in.scalar_type() == at::kFloat8_e5m2;
+    const bool full =
+        config().force_full_path && tr.is_gpu() &&
+        (is_bitwise(op) || is_pairloc(op) ||
+         (is_arith(op) &&
+          (native_reduce_dtype(tr, in.scalar_type()) || fp8_in)));
+    if (!full) return fast_clone(in);
+  }
+  if (is_pairloc(op)) {
+    return pairloc_lowered(tr, in, op);
   }
   if (is_logical(op)) {
     // land/lor/lxor lower to min/max/sum over 0/1 indicators; valid on any
@@ -345,15 +528,16 @@ Tensor allreduce_lowered(Transport& tr, const Tensor& in, int64_t op) {
       tr.allreduce(in, out, (RedOp)op);
       return out;
     }
-    // RCCL has no bitwise reductions: allgather + local CDNA4 reduce kernel.
-    auto staging = at::empty({tr.size() * in.numel()}, in.options());
-    tr.allgather_equal(in, staging);
-    auto out = at::empty_like(in);
+    // RCCL has no bitwise reductions: hierarchical exchange + local CDNA4
+    // reduce kernel (staging ~numel, not P*numel).
     const int kop = op == kBAnd ? 0 : (op == kBOr ? 1 : 2);
-    launch_bitwise_reduce(staging.data_ptr(), out.data_ptr(),
-                          in.numel() * in.element_size(), tr.size(), kop,
-                          current_gpu_stream(in));
-    return out;
+    const int64_t esize = in.element_size();
+    return hierarchical_allreduce(
+        tr, in, /*unit=*/1,
+        [&](const Tensor& stg, Tensor& blk, int64_t n, int nr) {
+          launch_bitwise_reduce(stg.data_ptr(), blk.data_ptr(), n * esize,
+                                nr, kop, current_gpu_stream(in));
+        });
   }
   TORCH_CHECK(is_arith(op));
   if (native_reduce_dtype(tr, in.scalar_type())) {
@@ -364,19 +548,19 @@ Tensor allreduce_lowered(Transport& tr, const Tensor& in, int64_t op) {
   const bool fp8 = in.scalar_type() == at::kFloat8_e4m3fn ||
                    in.scalar_type() == at::kFloat8_e5m2;
   if (fp8 && tr.is_gpu()) {
-    // fp8 stays fp8 on the wire: allgather the raw bytes, then one fused
-    // CDNA4 reduction with fp32 accumulators and a single quantization
-    // (better numerics than a per-hop-quantizing ring and 4x fewer wire
-    // bytes than the fp32-upcast path per hop).
-    auto staging = at::empty({tr.size() * in.numel()}, in.options());
-    tr.allgather_equal(in, staging);
-    auto out = at::empty_like(in);
+    // fp8 stays fp8 on the wire: hierarchical block exchange + one fused
+    // CDNA4 reduction per element with fp32 accumulators and a single
+    // quantization (better numerics than a per-hop-quantizing ring, 4x
+    // fewer wire bytes per hop than the fp32-upcast path, and — unlike
+    // the round-1 full allgather — staging stays ~numel at any P).
     const int kop = op == kSum ? 0 : (op == kProd ? 1 : (op == kMin ? 2 : 3));
-    launch_fp8_reduce(staging.data_ptr(), out.data_ptr(), in.numel(),
-                      tr.size(), kop,
-                      in.scalar_type() == at::kFloat8_e5m2,
-                      current_gpu_stream(in));
-    return out;
+    const bool e5m2 = in.scalar_type() == at::kFloat8_e5m2;
+    return hierarchical_allreduce(
+        tr, in, /*unit=*/1,
+        [&](const Tensor& stg, Tensor& blk, int64_t n, int nr) {
+          launch_fp8_reduce(stg.data_ptr(), blk.data_ptr(), n, nr, kop,
+                            e5m2, current_gpu_stream(in));
+        });
   }
   auto up = in.to(upcast_for_reduce(in.scalar_type()));
   auto red = at::empty_like(up);
@@ -461,6 +645,12 @@ struct AllreduceSumBackward : public M4ANode {
 
 Tensor Communicator::Allreduce(const Tensor& input, int64_t op) {
   check_op(op);
+  TORCH_CHECK(!is_pairloc(op) ||
+                  (input.dim() >= 1 && input.size(-1) == 2),
+              "mpi4torch_amd: MPI_MINLOC/MAXLOC operate on (value, "
+              "location) pairs — the last axis must have size 2 (the "
+              "tensor analog of MPI's pair types, reference "
+              "csrc/extension.cpp:204-252)");
   std::shared_ptr<M4ANode> grad_fn;
   if (torch::autograd::compute_requires_grad(input)) {
     if (op == kSum) {
@@ -546,6 +736,10 @@ struct ReduceSumInPlaceBackward : public M4ANode {
 
 Tensor Communicator::Reduce_(const Tensor& input, int64_t op, int64_t root) {
   check_op(op);
+  TORCH_CHECK(!is_pairloc(op) ||
+                  (input.dim() >= 1 && input.size(-1) == 2),
+              "mpi4torch_amd: MPI_MINLOC/MAXLOC operate on (value, "
+              "location) pairs — the last axis must have size 2");
   std::shared_ptr<M4ANode> grad_fn;
   if (torch::autograd::compute_requires_grad(input)) {
     if (op == kSum) {
@@ -725,7 +919,8 @@ Tensor Communicator::Gather(const Tensor& input, int64_t gatheraxis,
     auto in = stager.to_comm(input).contiguous().variable_data();
     auto& tr = tr_for(in);
     if (w1_shortcut(tr)) return stager.from_comm(fast_clone(in));
-    debug_check_collective(group_name_, "Gather", at::empty({0}), {gatheraxis, root});
+    debug_check_axis_collective(group_name_, "Gather", in, {gatheraxis},
+                                {gatheraxis, root});
     const auto g = axis_geom(in, gatheraxis);
     auto counts = host_allgather_int64(group_name_, g.axis);
     auto displs = prefix_displs(counts);
@@ -779,8 +974,8 @@ Tensor Communicator::Scatter(const Tensor& input, int64_t scatteraxis,
       return stager.from_comm(fast_clone(in));
     }
     const int me = tr.rank();
-    debug_check_collective(group_name_, "Scatter", at::empty({0}),
-                           {scatteraxis, root});
+    debug_check_dtype_collective(group_name_, "Scatter", in.scalar_type(),
+                                 {scatteraxis, root});
     // root broadcasts [ndim, sizes...]: the shape contract for non-root
     // ranks whose input tensor is a placeholder (reference :788-796)
     std::vector<int64_t> meta;
@@ -861,8 +1056,8 @@ Tensor Communicator::Allgather(const Tensor& input, int64_t gatheraxis) {
       }
       return stager.from_comm(fast_clone(in));
     }
-    debug_check_collective(group_name_, "Allgather", at::empty({0}),
-                           {gatheraxis});
+    debug_check_axis_collective(group_name_, "Allgather", in, {gatheraxis},
+                                {gatheraxis});
     const auto g = axis_geom(in, gatheraxis);
     auto counts = host_allgather_int64(group_name_, g.axis);
     auto displs = prefix_displs(counts);
@@ -1004,8 +1199,9 @@ Tensor Communicator::Reducescatter(const Tensor& input, int64_t axis,
                   "world size 1");
       return stager.from_comm(fast_clone(in));
     }
-    debug_check_collective(group_name_, "Reducescatter", at::empty({0}),
-                           {axis, numelem});
+    // numelem is per-rank (variable counts) — it must NOT enter the
+    // signature; the input tensor's full shape must match elementwise.
+    debug_check_collective(group_name_, "Reducescatter", in, {axis});
     const auto g = axis_geom(in, axis);
     auto counts = host_allgather_int64(group_name_, numelem);
     auto displs = prefix_displs(counts);
@@ -1125,8 +1321,9 @@ Tensor Communicator::AlltoallvImpl(const Tensor& input, int64_t gatheraxis,
     }
     const int me = tr.rank();
     const int P = tr.size();
-    debug_check_collective(group_name_, "Alltoall", at::empty({0}),
-                           {gatheraxis, scatteraxis});
+    debug_check_axis_collective(group_name_, "Alltoall", in,
+                                {gatheraxis, scatteraxis},
+                                {gatheraxis, scatteraxis, explicit_counts});
 
     // What each rank wants along the scatter axis. With explicit counts
     // (Alltoallv) both vectors are caller-provided and the two host
@@ -1311,6 +1508,8 @@ Tensor Communicator::AlltoallPairwise(const Tensor& input, int64_t axis,
       out.copy_(in);
       return stager.from_comm(std::move(out));
     }
+    debug_check_axis_collective(group_name_, "AlltoallPairwise", in, {axis},
+                                {axis});
     const auto g = axis_geom(in, axis);
     auto sblocks = make_blocks(in, g.before, g.after, send_counts, s_displs,
                                &in);
@@ -1623,6 +1822,26 @@ Tensor debug_fp8_reduce(const Tensor& stacked, int64_t op) {
                     (int)op, in.scalar_type() == at::kFloat8_e5m2,
                     current_gpu_stream(in));
   return out;
+}
+
+Tensor debug_pairloc_reduce(const Tensor& stacked, int64_t op) {
+  TORCH_CHECK(stacked.dim() >= 2 && stacked.size(-1) == 2,
+              "debug_pairloc_reduce expects [nranks, ..., 2] pairs");
+  auto in = stacked.contiguous();
+  const int64_t n = in.size(0);
+  auto out_sizes = in.sizes().slice(1).vec();
+  if (in.is_cuda()) {
+    auto out = at::empty(out_sizes, in.options());
+    launch_pairloc_reduce(in.data_ptr(), out.data_ptr(), out.numel() / 2,
+                          (int)n, (int)op,
+                          pairloc_dtype_code(in.scalar_type()),
+                          current_gpu_stream(in));
+    return out;
+  }
+  // CPU reference composite — what the CDNA4 kernel is compared against
+  return pairloc_local_reduce(in.view({n, -1, 2}), op, in.scalar_type())
+      .view(out_sizes)
+      .contiguous();
 }
 
 Tensor debug_bitwise_reduce(const Tensor& stacked, int64_t op) {

@@ -117,5 +117,7 @@ at::Tensor debug_pack_roundtrip(const at::Tensor& input, int64_t axis,
                                 std::vector<int64_t> counts);
 at::Tensor debug_bitwise_reduce(const at::Tensor& stacked, int64_t op);
 at::Tensor debug_fp8_reduce(const at::Tensor& stacked, int64_t op);
+// stacked: [nranks, ..., 2] (value, location) pairs; op: 0=minloc 1=maxloc
+at::Tensor debug_pairloc_reduce(const at::Tensor& stacked, int64_t op);
 
 } // namespace m4a

@@ -113,12 +113,12 @@ def _errors_worker(rank, world):
         raise AssertionError("expected RuntimeError for MAX backward")
     except RuntimeError:
         pass
-    # MINLOC/MAXLOC have no tensor equivalent
+    # MINLOC/MAXLOC require (value, location) pairs: last axis size 2
     try:
-        comm.Allreduce(torch.rand(4), m.MPI_MINLOC)
-        raise AssertionError("expected RuntimeError for MINLOC")
-    except RuntimeError:
-        pass
+        comm.Allreduce(torch.rand(3), m.MPI_MINLOC)
+        raise AssertionError("expected RuntimeError for bad MINLOC shape")
+    except RuntimeError as e:
+        assert "pairs" in str(e)
 
 
 def test_allreduce_sum_ws2():
@@ -139,6 +139,72 @@ def test_allreduce_ops_ws7():
 
 def test_allreduce_dtypes_ws2():
     run_spmd(2, _dtypes_worker)
+
+
+def _pairloc_worker(rank, world):
+    import mpi4torch_amd as m
+
+    comm = m.COMM_WORLD
+    # (value, location) pairs along the last axis, MPI pair-type semantics
+    # (reference op table csrc/extension.cpp:204-252). Values are arranged
+    # so every rank can compute the expected winner in closed form.
+    n = 11
+    vals = torch.tensor(
+        [((i * 7 + rank * 3) % world) - (world / 2.0) for i in range(n)]
+    )
+    locs = torch.full((n,), float(rank * 100 + 5))
+    pairs = torch.stack([vals, locs], dim=-1)
+
+    got_min = comm.Allreduce(pairs, m.MPI_MINLOC)
+    got_max = comm.Allreduce(pairs, m.MPI_MAXLOC)
+
+    # reference computation on the allgathered pairs
+    allv = comm.Allgather(vals.reshape(1, n), 0)
+    alll = comm.Allgather(locs.reshape(1, n), 0)
+    for i in range(n):
+        col = allv[:, i]
+        for got, extreme in ((got_min, col.min()), (got_max, col.max())):
+            mask = col == extreme
+            want_loc = alll[:, i][mask].min()
+            assert got[i, 0] == extreme, (i, got[i], extreme)
+            assert got[i, 1] == want_loc, (i, got[i], want_loc)
+
+    # ties: identical values everywhere -> smallest location wins
+    tie = torch.stack(
+        [torch.ones(5), torch.full((5,), float(world - rank))], dim=-1
+    )
+    r = comm.Allreduce(tie, m.MPI_MINLOC)
+    assert (r[:, 0] == 1).all()
+    assert (r[:, 1] == 1).all()  # smallest location = world - (world-1)
+
+    # MINLOC/MAXLOC backward is unimplemented (like all non-SUM ops)
+    p = pairs.clone().requires_grad_()
+    res = comm.Allreduce(p, m.MPI_MAXLOC)
+    try:
+        res.sum().backward()
+        raise AssertionError("expected RuntimeError for MINLOC backward")
+    except RuntimeError:
+        pass
+
+    # integer dtype pairs
+    ip = torch.stack(
+        [torch.arange(4, dtype=torch.int64) * (rank + 1),
+         torch.full((4,), rank, dtype=torch.int64)], dim=-1
+    )
+    ri = comm.Allreduce(ip, m.MPI_MAXLOC)
+    assert ri.dtype == torch.int64
+    assert (ri[:, 0] == torch.arange(4) * world).all()
+    if world > 1:
+        assert ri[0, 1] == 0  # all ranks tie at value 0 -> smallest loc
+        assert (ri[1:, 1] == world - 1).all()
+
+
+def test_allreduce_pairloc_ws2():
+    run_spmd(2, _pairloc_worker)
+
+
+def test_allreduce_pairloc_ws5():
+    run_spmd(5, _pairloc_worker)
 
 
 def test_allreduce_errors_ws2():

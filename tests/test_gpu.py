@@ -210,6 +210,64 @@ def test_fp8_allreduce_path(world1):
         assert (r2.view(torch.uint8) == t.view(torch.uint8)).all()
 
 
+def test_pairloc_reduce_kernel(world1):
+    m = world1
+    # CDNA4 pair arg-reduce vs the torch-composite reference (same function
+    # the CPU transport path uses)
+    for dtype in (torch.float32, torch.float64, torch.float16,
+                  torch.bfloat16, torch.int32, torch.int64, torch.int16,
+                  torch.int8, torch.uint8):
+        for n in (1024, 1027):
+            if dtype.is_floating_point:
+                vals = (torch.randn(6, n) * 4).to(dtype)
+            else:
+                vals = torch.randint(0, 50, (6, n)).to(dtype)
+            locs = torch.randint(0, 90, (6, n)).to(dtype)
+            stacked = torch.stack([vals, locs], dim=-1)
+            for op in (0, 1):  # minloc, maxloc
+                ref = m._C._pairloc_reduce(stacked, op)  # CPU composite
+                got = m._C._pairloc_reduce(stacked.cuda(), op).cpu()
+                torch.cuda.synchronize()
+                assert (got == ref).all(), (dtype, n, op)
+
+
+def test_pairloc_allreduce_full_path(world1):
+    m = world1
+    comm = m.COMM_WORLD
+    # world-1 force_full_path drives the whole hierarchical machinery
+    # (self block exchange -> arg-reduce kernel -> in-place allgather)
+    m._C.force_full_path(True)
+    try:
+        pairs = torch.stack(
+            [torch.randn(4097, device="cuda"),
+             torch.randint(0, 100, (4097,), device="cuda").float()], dim=-1)
+        for op in (m.MPI_MINLOC, m.MPI_MAXLOC):
+            r = comm.Allreduce(pairs, op)
+            torch.testing.assert_close(r, pairs)  # P=1: identity
+    finally:
+        m._C.force_full_path(False)
+
+
+def test_hierarchical_fp8_bitwise_full_path(world1):
+    m = world1
+    comm = m.COMM_WORLD
+    # the hierarchical (exchange + fused kernel + in-place allgather) paths
+    # at P=1 must be exact identities, including non-divisible sizes
+    m._C.force_full_path(True)
+    try:
+        for n in (1 << 16, (1 << 16) + 3):
+            t8 = (torch.randn(n, device="cuda") * 0.5).to(torch.float8_e4m3fn)
+            r8 = comm.Allreduce(t8, m.MPI_SUM)
+            assert (r8.view(torch.uint8) == t8.view(torch.uint8)).all()
+            ti = torch.randint(0, 1 << 30, (n,), device="cuda",
+                               dtype=torch.int32)
+            for op in (m.MPI_BAND, m.MPI_BOR, m.MPI_BXOR):
+                ri = comm.Allreduce(ti, op)
+                assert (ri == ti).all()
+    finally:
+        m._C.force_full_path(False)
+
+
 def test_hipgraph_capture(world1):
     m = world1
     comm = m.COMM_WORLD

@@ -55,6 +55,7 @@ static auto communicator_class =
         .def("AlltoallPairwise", &Communicator::AlltoallPairwise)
         .def("Iallreduce", &Communicator::Iallreduce)
         .def("Ireducescatter", &Communicator::Ireducescatter)
+        .def("Iallgather", &Communicator::Iallgather)
         .def("Isend", &Communicator::Isend)
         .def("Irecv", &Communicator::Irecv)
         .def("Wait", &Communicator::Wait)

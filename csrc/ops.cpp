@@ -1733,6 +1733,29 @@ std::vector<at::Tensor> Communicator::Ireducescatter(const Tensor& input,
   return {meta, out, in};
 }
 
+std::vector<at::Tensor> Communicator::Iallgather(const Tensor& input) {
+  TORCH_CHECK(!torch::autograd::compute_requires_grad(input),
+              "mpi4torch_amd: Iallgather does not support autograd; use "
+              "Allgather, or detach the input (FSDP parameter prefetch "
+              "operates on detached shards)");
+  at::AutoDispatchBelowADInplaceOrView guard;
+  DeviceStager stager(input);
+  auto in = stager.to_comm(input).contiguous().variable_data();
+  auto& tr = tr_for(in);
+  auto out = at::empty({(int64_t)tr.size() * in.numel()}, in.options());
+  uint64_t req = tr.iallgather(in, out);
+  auto meta = at::empty({7}, at::TensorOptions().dtype(at::kDouble));
+  auto* m = meta.data_ptr<double>();
+  m[0] = (double)req;
+  m[1] = (double)kIallreduceOp;  // same wait semantics: handle returns buffer
+  m[2] = 0.0;
+  m[3] = 0.0;
+  m[4] = ptr_hash(out.data_ptr());
+  m[5] = (double)(int64_t)stager.orig_device_.type();
+  m[6] = (double)stager.orig_device_.index();
+  return {meta, out, in};
+}
+
 std::vector<at::Tensor> Communicator::Isend(const Tensor& input, int64_t dest,
                                             int64_t tag) {
   return IsendImpl(input, dest, tag, /*backward_channel=*/false);

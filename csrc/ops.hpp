@@ -75,6 +75,9 @@ struct Communicator : torch::CustomClassHolder {
   // flat block of the elementwise sum. input.numel() must be
   // world_size * block; returns a wait handle for the [block] result.
   std::vector<at::Tensor> Ireducescatter(const at::Tensor& input, int64_t op);
+  // Non-blocking equal-count flat allgather (no autograd): Wait yields the
+  // rank-major concatenation [size * numel]. FSDP prefetch primitive.
+  std::vector<at::Tensor> Iallgather(const at::Tensor& input);
 
   // Non-blocking p2p. Handle contract identical to the reference
   // (csrc/extension.cpp:1094-1107): [meta tensor, comm buffer, input].

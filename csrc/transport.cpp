@@ -419,6 +419,10 @@ struct LocalTransport final : Transport {
                            RedOp op) override {
     return iallreduce(in.view_as(out), out, op);
   }
+  uint64_t iallgather(const at::Tensor& in, at::Tensor& out) override {
+    auto v = out.view_as(in);  // size 1: out and in have equal numel
+    return iallreduce(in, v, kSum);
+  }
   uint64_t iallreduce(const at::Tensor& in, at::Tensor& out,
                       RedOp) override {
     out.copy_(in, true);
@@ -575,6 +579,18 @@ struct C10dTransport final : Transport {
     reduce_scatter_equal(in, out, op);
     Request r;
     r.buffer = out;
+    return requests().add(std::move(r));
+  }
+  uint64_t iallgather(const at::Tensor& in, at::Tensor& out) override {
+    std::lock_guard<std::mutex> g(mu_);
+    auto chunks = out.view({size(), -1}).unbind(0);
+    std::vector<at::Tensor> outs;
+    for (auto& c : chunks) outs.push_back(c.view_as(in));
+    std::vector<std::vector<at::Tensor>> outputs{outs};
+    std::vector<at::Tensor> inputs{const_cast<at::Tensor&>(in)};
+    Request r;
+    r.buffer = out;
+    r.work = backend_->allgather(outputs, inputs);
     return requests().add(std::move(r));
   }
 
@@ -745,6 +761,15 @@ struct RcclTransport final : Transport {
   void flush_p2p() override {
     std::lock_guard<std::mutex> g(mu_);
     flush_pending_locked();
+  }
+  uint64_t iallgather(const at::Tensor& in, at::Tensor& out) override {
+    std::lock_guard<std::mutex> g(mu_);
+    check_not_capturing("Iallgather");
+    enter_side(Channel::Coll, {in, out});
+    M4A_NCCL_CHECK(ncclAllGather(in.data_ptr(), out.data_ptr(), nbytes(in),
+                                 ncclUint8, comm(Channel::Coll),
+                                 stream(Channel::Coll)));
+    return make_gpu_request(Channel::Coll, out);
   }
   uint64_t ireduce_scatter(const at::Tensor& in, at::Tensor& out,
                            RedOp op) override {

@@ -90,6 +90,9 @@ struct Transport : std::enable_shared_from_this<Transport> {
   // primitive. in.numel() == size * out.numel(), rank-major blocks.
   virtual uint64_t ireduce_scatter(const at::Tensor& in, at::Tensor& out,
                                    RedOp op) = 0;
+  // Non-blocking equal-count allgather (no autograd): the FSDP parameter
+  // prefetch primitive. out.numel() == size * in.numel(), rank-major.
+  virtual uint64_t iallgather(const at::Tensor& in, at::Tensor& out) = 0;
   // Whether fp8 reductions run natively (RCCL probes at runtime; the cast
   // fallback is used otherwise). Non-GPU transports: false.
   virtual bool fp8_reduce_supported(at::ScalarType) { return false; }

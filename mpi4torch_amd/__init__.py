@@ -242,6 +242,12 @@ class MPI_Communicator:
         The ZeRO-2 bucket primitive; not in the reference API."""
         return WaitHandle(self._comm.Ireducescatter(tensor, op))
 
+    def Iallgather(self, tensor: torch.Tensor) -> WaitHandle:
+        """Non-blocking equal-count flat allgather (no autograd): Wait()
+        yields the rank-major concatenation (size * numel). The FSDP
+        parameter-prefetch primitive; not in the reference API."""
+        return WaitHandle(self._comm.Iallgather(tensor))
+
     def Isend(self, tensor: torch.Tensor, dest: int, tag: int) -> WaitHandle:
         """Non-blocking send; complete with Wait. Backward: reverse recv.
 

@@ -60,6 +60,7 @@ class _Unit:
                 off += n
         self.materialized = True
         self.grad_handle = None
+        self.prefetch_handle = None
         self.pending = 0
 
     @torch.no_grad()
@@ -69,11 +70,30 @@ class _Unit:
             self.materialized = False
 
     @torch.no_grad()
+    def start_materialize(self):
+        """Prefetch: launch this unit's parameter allgather without waiting
+        (Iallgather skips the returning stream bracket), so it overlaps the
+        CURRENT unit's compute. materialize() consumes the handle."""
+        if (self.materialized or self.prefetch_handle is not None
+                or self.comm.size == 1):
+            return
+        local = self.shard.detach()
+        if local.dtype != self.flat.dtype:
+            local = local.to(self.flat.dtype)
+        self.prefetch_handle = self.comm.Iallgather(local)
+
+    @torch.no_grad()
     def materialize(self):
         if self.materialized:
             return
         self.flat.untyped_storage().resize_(
             self.padded * self.flat.element_size())
+        if self.prefetch_handle is not None:
+            full = self.comm.Wait(self.prefetch_handle)
+            self.prefetch_handle = None
+            self.flat.copy_(full)
+            self.materialized = True
+            return
         local = self.shard.detach()
         if local.dtype != self.flat.dtype:
             local = local.to(self.flat.dtype)  # model dtype on the wire
@@ -152,7 +172,13 @@ class FullyShardedDataParallel(torch.nn.Module):
 
     # ---- hooks -----------------------------------------------------------
     def _pre_forward(self, module, inputs):
-        self._by_module[module].materialize()
+        u = self._by_module[module]
+        u.materialize()
+        # prefetch the NEXT unit's allgather so it overlaps this unit's
+        # forward compute (unit order = construction order)
+        i = self._units.index(u)
+        if i + 1 < len(self._units):
+            self._units[i + 1].start_materialize()
 
     def _post_forward(self, module, inputs, output):
         u = self._by_module[module]
@@ -161,7 +187,12 @@ class FullyShardedDataParallel(torch.nn.Module):
         return output
 
     def _pre_backward(self, module, grad_output):
-        self._by_module[module].materialize()
+        u = self._by_module[module]
+        u.materialize()
+        # backward visits units in reverse: prefetch the PREVIOUS unit
+        i = self._units.index(u)
+        if i > 0:
+            self._units[i - 1].start_materialize()
 
     def _grad_ready(self, p):
         if not self._sync_enabled:

@@ -511,6 +511,30 @@ def test_fsdp_accumulation_ws2():
     run_spmd(2, _fsdp_accum_worker)
 
 
+def _iallgather_worker(rank, world):
+    import mpi4torch_amd as m
+    from mpi4torch_amd.parallel import FullyShardedDataParallel  # noqa: F401
+
+    comm = m.COMM_WORLD
+    t = torch.full((5,), float(rank + 1), dtype=torch.float64)
+    h = comm.Iallgather(t)
+    got = comm.Wait(h)
+    assert got.shape == (world * 5,)
+    for r in range(world):
+        assert (got[r * 5 : (r + 1) * 5] == r + 1).all()
+    # two in flight at once (the FSDP prefetch pattern)
+    h1 = comm.Iallgather(torch.full((3,), float(rank)))
+    h2 = comm.Iallgather(torch.full((2,), float(rank * 10)))
+    g1, g2 = comm.Wait(h1), comm.Wait(h2)
+    for r in range(world):
+        assert (g1[r * 3 : (r + 1) * 3] == r).all()
+        assert (g2[r * 2 : (r + 1) * 2] == r * 10).all()
+
+
+def test_iallgather_ws3():
+    run_spmd(3, _iallgather_worker)
+
+
 class _BranchyNet(torch.nn.Module):
     """Shared trunk + a branch only some ranks execute."""
 

@@ -81,8 +81,17 @@ struct Config {
   // indefinitely (default). (GPU-side progress is stream-ordered; use the
   // desync detector for pre-enqueue divergence.)
   int64_t op_timeout_ms = 0;
+  // MPI4TORCH_AMD_PIPELINE_MB (float, default 64): chunk size for the
+  // phased pack->wire pipelining of axis-marshaling collectives. Payloads
+  // whose packed side exceeds one chunk are exchanged in up to 4 phases so
+  // the CDNA4 pack/unpack kernels overlap the wire time of neighboring
+  // phases. <=0 disables phasing.
+  int64_t pipeline_chunk_bytes = 64ll << 20;
 };
 
 Config& config();
+// Re-read the MPI4TORCH_AMD_* environment into the live config (tests and
+// long-lived processes that toggle behavior between phases).
+void reload_config_from_env();
 
 } // namespace m4a

@@ -104,6 +104,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("_bitwise_reduce", &m4a::debug_bitwise_reduce);
   m.def("_fp8_reduce", &m4a::debug_fp8_reduce);
   m.def("_pairloc_reduce", &m4a::debug_pairloc_reduce);
+  m.def("reload_config", &m4a::reload_config_from_env);
 
   m.def("_rccl_version", []() {
     int v = 0;

@@ -137,12 +137,18 @@ hipStream_t current_gpu_stream(const Tensor& t) {
 // `blocks` (pack=true) or back (pack=false). GPU: one batched CDNA4 slab
 // kernel launch on the current stream; CPU: strided tensor copies. This is
 // the MI355X-native replacement for MPI_Type_vector marshaling (reference
-// csrc/extension.cpp:556-577).
+// csrc/extension.cpp:556-577). The optional [row_lo, row_hi) range limits
+// the copy to a slice of the `before` dimension — one phase of the
+// pipelined (chunked) exchange; defaults cover all rows.
 void move_axis_blocks(const Tensor& full, int64_t axis,
                       const std::vector<int64_t>& displs,
                       const std::vector<int64_t>& counts,
-                      std::vector<Tensor>& blocks, bool pack) {
+                      std::vector<Tensor>& blocks, bool pack,
+                      int64_t row_lo = 0, int64_t row_hi = -1) {
   const auto g = axis_geom(full, axis);
+  if (row_hi < 0) row_hi = g.before;
+  const int64_t rows = row_hi - row_lo;
+  if (rows <= 0) return;
   if (full.is_cuda()) {
     std::vector<SlabDesc> descs;
     descs.reserve(counts.size());
@@ -151,18 +157,18 @@ void move_axis_blocks(const Tensor& full, int64_t axis,
       if (counts[i] == 0 || g.before * g.after == 0) continue;
       char* bp = static_cast<char*>(blocks[i].data_ptr());
       SlabDesc d;
-      d.before = g.before;
+      d.before = rows;
       d.count = counts[i];
       d.after_b = g.after_b;
       if (pack) {
-        d.src = base + displs[i] * g.after_b;
+        d.src = base + displs[i] * g.after_b + row_lo * g.axis * g.after_b;
         d.src_pitch_b = g.axis * g.after_b;
-        d.dst = bp;
+        d.dst = bp + row_lo * counts[i] * g.after_b;
         d.dst_pitch_b = counts[i] * g.after_b;
       } else {
-        d.src = bp;
+        d.src = bp + row_lo * counts[i] * g.after_b;
         d.src_pitch_b = counts[i] * g.after_b;
-        d.dst = base + displs[i] * g.after_b;
+        d.dst = base + displs[i] * g.after_b + row_lo * g.axis * g.after_b;
         d.dst_pitch_b = g.axis * g.after_b;
       }
       descs.push_back(d);
@@ -172,15 +178,17 @@ void move_axis_blocks(const Tensor& full, int64_t axis,
                        current_gpu_stream(full));
     }
   } else {
-    auto shape = full.sizes().vec();
+    if (g.before * g.after == 0) return;
+    auto f3 = full.view({g.before, g.axis, g.after})
+                  .narrow(0, row_lo, rows);
     for (size_t i = 0; i < counts.size(); ++i) {
       if (counts[i] == 0) continue;
-      shape[axis] = counts[i];
-      auto view = blocks[i].view(shape);
+      auto b3 = blocks[i].view({g.before, counts[i], g.after})
+                    .narrow(0, row_lo, rows);
       if (pack) {
-        view.copy_(full.narrow(axis, displs[i], counts[i]));
+        b3.copy_(f3.narrow(1, displs[i], counts[i]));
       } else {
-        full.narrow(axis, displs[i], counts[i]).copy_(view);
+        f3.narrow(1, displs[i], counts[i]).copy_(b3);
       }
     }
   }
@@ -1345,6 +1353,7 @@ Tensor Communicator::AlltoallvImpl(const Tensor& input, int64_t gatheraxis,
     std::vector<int64_t> r_offs(P), r_lens(P);  // my recv slices (in `out`)
     std::vector<int64_t> outsizes = in.sizes().vec();
     int64_t send_axis = scatteraxis, recv_axis = gatheraxis;
+    std::vector<int64_t> srcsz;  // per-rank source-axis sizes (shared)
 
     if (gatheraxis != scatteraxis) {
       const int64_t stotal = sdispls.back() + scounts.back();
@@ -1367,6 +1376,7 @@ Tensor Communicator::AlltoallvImpl(const Tensor& input, int64_t gatheraxis,
       }
       outsizes[gatheraxis] = gdispls.back() + gsizes.back();
       outsizes[scatteraxis] = scounts[me];
+      srcsz = gsizes;
     } else {
       // same-axis repartition (reference :947-979): ranks hold chunks of a
       // global axis; redistribute to the partition given by numelem
@@ -1398,19 +1408,117 @@ Tensor Communicator::AlltoallvImpl(const Tensor& input, int64_t gatheraxis,
         r_lens[j] = std::max<int64_t>(0, rhi - rlo);
       }
       outsizes[gatheraxis] = scounts[me];
+      srcsz = nsizes;
     }
 
     auto out = at::empty(outsizes, in.options());
     const auto gs = axis_geom(in, send_axis);
-    auto sblocks = make_blocks(in, gs.before, gs.after, s_lens, s_offs, &in);
-    if (gs.before != 1) {
-      move_axis_blocks(in, send_axis, s_offs, s_lens, sblocks, /*pack=*/true);
-    }
     const auto gr = axis_geom(out, recv_axis);
+    const bool send_packed = gs.before != 1;
+    const bool recv_packed = gr.before != 1;
+    auto sblocks = make_blocks(in, gs.before, gs.after, s_lens, s_offs, &in);
     auto rblocks = make_blocks(out, gr.before, gr.after, r_lens, r_offs, &out);
     auto peers = iota_peers(P);
+
+    // ---- chunked pack -> wire pipelining (VERDICT/TODO task: overlap the
+    // CDNA4 marshaling kernels with the wire). The marshaled side's slab
+    // rows are split into K phases; each phase is its own full-width
+    // grouped exchange (all peers, all links — no loss of multi-link
+    // concurrency) issued non-blockingly, so pack(k+1) and unpack(k) run
+    // on the compute stream while phase k rides the collective stream.
+    // K is derived from SHARED quantities only (global logical tensor
+    // size, common non-axis dims), so every rank slices every pair block
+    // at identical byte offsets. Supported: exactly one marshaled side
+    // (different axes), or both (same axis — rows coincide); the rare
+    // both-marshaled different-axis case falls back to one phase.
+    int64_t K = 1;
+    if (config().pipeline_chunk_bytes > 0 && (send_packed || recv_packed) &&
+        (send_axis == recv_axis || send_packed != recv_packed)) {
+      int64_t global_elems = 1;
+      for (size_t d = 0; d < outsizes.size(); ++d) {
+        global_elems *= (d == (size_t)scatteraxis)
+                            ? (sdispls.back() + scounts.back())
+                            : outsizes[d];
+      }
+      const int64_t per_rank_b = global_elems / P * in.element_size();
+      K = std::min<int64_t>(
+          4, std::max<int64_t>(
+                 1, (per_rank_b + config().pipeline_chunk_bytes - 1) /
+                        config().pipeline_chunk_bytes));
+    }
+
+    if (K > 1) {
+      const int me_ = me;
+      // per-pair row geometry of the marshaled side, from shared data
+      auto send_geom = [&](int j) -> std::pair<int64_t, int64_t> {
+        if (send_axis == recv_axis || send_packed) {
+          return {gs.before, s_lens[j] * gs.after};
+        }
+        // recv side is the marshaled one: receiver j's rows (its out dims
+        // = common outsizes with the scatter axis at scounts[j])
+        int64_t before = 1, after = 1;
+        for (int64_t d = 0; d < (int64_t)outsizes.size(); ++d) {
+          const int64_t sz =
+              (d == scatteraxis) ? scounts[j] : outsizes[d];
+          if (d < recv_axis) before *= sz;
+          else if (d > recv_axis) after *= sz;
+        }
+        return {before, srcsz[me_] * after};
+      };
+      auto recv_geom = [&](int i) -> std::pair<int64_t, int64_t> {
+        if (send_axis == recv_axis || recv_packed) {
+          return {gr.before, r_lens[i] * gr.after};
+        }
+        // send side is the marshaled one: sender i's rows (its in dims =
+        // mine with the gather axis at srcsz[i])
+        int64_t before = 1, after = 1;
+        auto insz = in.sizes();
+        for (int64_t d = 0; d < (int64_t)insz.size(); ++d) {
+          const int64_t sz = (d == recv_axis) ? srcsz[i] : insz[d];
+          if (d < send_axis) before *= sz;
+          else if (d > send_axis) after *= sz;
+        }
+        return {before, scounts[me_] * after};
+      };
+      auto bound = [&](int64_t rows, int64_t k) { return rows * k / K; };
+      auto slice = [](const Tensor& blk, int64_t lo, int64_t len) {
+        auto flat = blk.view({-1});
+        return len > 0 ? flat.narrow(0, lo, len) : flat.narrow(0, 0, 0);
+      };
+      std::vector<uint64_t> reqs((size_t)K);
+      for (int64_t k = 0; k < K; ++k) {
+        if (send_packed) {
+          move_axis_blocks(in, send_axis, s_offs, s_lens, sblocks, true,
+                           bound(gs.before, k), bound(gs.before, k + 1));
+        }
+        std::vector<Tensor> sk(P), rk(P);
+        for (int j = 0; j < P; ++j) {
+          auto sg = send_geom(j);
+          sk[j] = slice(sblocks[j], bound(sg.first, k) * sg.second,
+                        (bound(sg.first, k + 1) - bound(sg.first, k)) *
+                            sg.second);
+          auto rg = recv_geom(j);
+          rk[j] = slice(rblocks[j], bound(rg.first, k) * rg.second,
+                        (bound(rg.first, k + 1) - bound(rg.first, k)) *
+                            rg.second);
+        }
+        reqs[(size_t)k] = tr.iexchange(sk, peers, rk, peers);
+      }
+      for (int64_t k = 0; k < K; ++k) {
+        wait_request(reqs[(size_t)k]);
+        if (recv_packed) {
+          move_axis_blocks(out, recv_axis, r_offs, r_lens, rblocks, false,
+                           bound(gr.before, k), bound(gr.before, k + 1));
+        }
+      }
+      return stager.from_comm(std::move(out));
+    }
+
+    if (send_packed) {
+      move_axis_blocks(in, send_axis, s_offs, s_lens, sblocks, /*pack=*/true);
+    }
     tr.exchange(sblocks, peers, rblocks, peers);
-    if (gr.before != 1) {
+    if (recv_packed) {
       move_axis_blocks(out, recv_axis, r_offs, r_lens, rblocks,
                        /*pack=*/false);
     }

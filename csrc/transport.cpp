@@ -27,20 +27,29 @@
 
 namespace m4a {
 
+static void config_from_env(Config& c) {
+  const char* dbg = std::getenv("MPI4TORCH_AMD_DEBUG");
+  c.debug_collectives = dbg && dbg[0] == '1';
+  const char* ffp = std::getenv("MPI4TORCH_AMD_FORCE_FULL_PATH");
+  c.force_full_path = ffp && ffp[0] == '1';
+  if (const char* to = std::getenv("MPI4TORCH_AMD_TIMEOUT_S")) {
+    c.op_timeout_ms = (int64_t)(std::atof(to) * 1000.0);
+  }
+  if (const char* pc = std::getenv("MPI4TORCH_AMD_PIPELINE_MB")) {
+    c.pipeline_chunk_bytes = (int64_t)(std::atof(pc) * 1048576.0);
+  }
+}
+
 Config& config() {
   static Config cfg = []() {
     Config c;
-    const char* dbg = std::getenv("MPI4TORCH_AMD_DEBUG");
-    c.debug_collectives = dbg && dbg[0] == '1';
-    const char* ffp = std::getenv("MPI4TORCH_AMD_FORCE_FULL_PATH");
-    c.force_full_path = ffp && ffp[0] == '1';
-    if (const char* to = std::getenv("MPI4TORCH_AMD_TIMEOUT_S")) {
-      c.op_timeout_ms = (int64_t)(std::atof(to) * 1000.0);
-    }
+    config_from_env(c);
     return c;
   }();
   return cfg;
 }
+
+void reload_config_from_env() { config_from_env(config()); }
 
 #define M4A_HIP_CHECK(expr)                                              \
   do {                                                                   \
@@ -77,7 +86,9 @@ struct Request {
   hipEvent_t event = nullptr;
   int device = -1;
   c10::intrusive_ptr<c10d::Work> work;
+  std::vector<c10::intrusive_ptr<c10d::Work>> works;  // multi-op requests
   at::Tensor buffer;  // keeps the comm buffer alive until wait
+  std::vector<at::Tensor> buffers;  // multi-op requests
   // MPI4TORCH_AMD_DEBUG=1 p2p handshake: the sender ships [tag, dtype,
   // numel] over the host channel; the receiver compares at Wait and raises
   // on mismatch — FIFO-crossed transfers become errors, not silent swaps.
@@ -406,6 +417,22 @@ struct LocalTransport final : Transport {
       recvbufs[i].view({-1}).copy_(sendbufs[i].reshape({-1}), true);
     }
   }
+  uint64_t iexchange(const std::vector<at::Tensor>& sendbufs,
+                     const std::vector<int>& speers,
+                     std::vector<at::Tensor>& recvbufs,
+                     const std::vector<int>& rpeers) override {
+    exchange(sendbufs, speers, recvbufs, rpeers);
+    Request r;
+    r.buffers = recvbufs;
+    if (!recvbufs.empty() && recvbufs[0].is_cuda()) {
+      r.gpu = true;
+      r.device = (int)recvbufs[0].get_device();
+      auto cur = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(r.device);
+      r.event = EventPool::forDevice(r.device).acquire();
+      M4A_HIP_CHECK(hipEventRecord(r.event, cur.stream()));
+    }
+    return requests().add(std::move(r));
+  }
   uint64_t isend(const at::Tensor& buf, int peer, int tag,
                  Channel ch) override {
     TORCH_CHECK(peer == 0, "world_size is 1; cannot send to rank ", peer);
@@ -534,6 +561,42 @@ struct C10dTransport final : Transport {
       self_recvs[i]->view({-1}).copy_(self_sends[i]->reshape({-1}));
     }
     for (auto& w : works) wait_work(w);
+  }
+  uint64_t iexchange(const std::vector<at::Tensor>& sendbufs,
+                     const std::vector<int>& speers,
+                     std::vector<at::Tensor>& recvbufs,
+                     const std::vector<int>& rpeers) override {
+    std::lock_guard<std::mutex> g(mu_);
+    const int me = rank();
+    Request r;
+    std::vector<const at::Tensor*> self_sends;
+    std::vector<at::Tensor*> self_recvs;
+    for (size_t j = 0; j < recvbufs.size(); ++j) {
+      if (rpeers[j] == me) {
+        self_recvs.push_back(&recvbufs[j]);
+        continue;
+      }
+      if (recvbufs[j].numel() == 0) continue;
+      std::vector<at::Tensor> ts{recvbufs[j]};
+      r.works.push_back(backend_->recv(ts, rpeers[j], /*tag=*/0));
+      r.buffers.push_back(recvbufs[j]);
+    }
+    for (size_t i = 0; i < sendbufs.size(); ++i) {
+      if (speers[i] == me) {
+        self_sends.push_back(&sendbufs[i]);
+        continue;
+      }
+      if (sendbufs[i].numel() == 0) continue;
+      std::vector<at::Tensor> ts{const_cast<at::Tensor&>(sendbufs[i])};
+      r.works.push_back(backend_->send(ts, speers[i], /*tag=*/0));
+      r.buffers.push_back(sendbufs[i]);
+    }
+    TORCH_CHECK(self_sends.size() == self_recvs.size(),
+                "iexchange: unmatched self blocks");
+    for (size_t i = 0; i < self_sends.size(); ++i) {
+      self_recvs[i]->view({-1}).copy_(self_sends[i]->reshape({-1}));
+    }
+    return requests().add(std::move(r));
   }
   // p2p matching is FIFO per (peer, channel) — the fixed per-channel gloo
   // tag reproduces the RCCL transport's tag-free semantics exactly, so the
@@ -761,6 +824,57 @@ struct RcclTransport final : Transport {
   void flush_p2p() override {
     std::lock_guard<std::mutex> g(mu_);
     flush_pending_locked();
+  }
+  uint64_t iexchange(const std::vector<at::Tensor>& sendbufs,
+                     const std::vector<int>& speers,
+                     std::vector<at::Tensor>& recvbufs,
+                     const std::vector<int>& rpeers) override {
+    std::lock_guard<std::mutex> g(mu_);
+    check_not_capturing("iexchange");
+    std::vector<at::Tensor> all(sendbufs);
+    all.insert(all.end(), recvbufs.begin(), recvbufs.end());
+    enter_side(Channel::Coll, all);
+    const int me = rank_;
+    std::vector<const at::Tensor*> self_sends;
+    std::vector<at::Tensor*> self_recvs;
+    M4A_NCCL_CHECK(ncclGroupStart());
+    for (size_t i = 0; i < sendbufs.size(); ++i) {
+      if (speers[i] == me) {
+        self_sends.push_back(&sendbufs[i]);
+        continue;
+      }
+      if (nbytes(sendbufs[i]) == 0) continue;
+      M4A_NCCL_CHECK(ncclSend(sendbufs[i].data_ptr(), nbytes(sendbufs[i]),
+                              ncclUint8, speers[i], comm(Channel::Coll),
+                              stream(Channel::Coll)));
+    }
+    for (size_t j = 0; j < recvbufs.size(); ++j) {
+      if (rpeers[j] == me) {
+        self_recvs.push_back(&recvbufs[j]);
+        continue;
+      }
+      if (nbytes(recvbufs[j]) == 0) continue;
+      M4A_NCCL_CHECK(ncclRecv(recvbufs[j].data_ptr(), nbytes(recvbufs[j]),
+                              ncclUint8, rpeers[j], comm(Channel::Coll),
+                              stream(Channel::Coll)));
+    }
+    M4A_NCCL_CHECK(ncclGroupEnd());
+    TORCH_CHECK(self_sends.size() == self_recvs.size(),
+                "iexchange: unmatched self blocks");
+    for (size_t i = 0; i < self_sends.size(); ++i) {
+      M4A_HIP_CHECK(hipMemcpyAsync(
+          self_recvs[i]->data_ptr(), self_sends[i]->data_ptr(),
+          nbytes(*self_sends[i]), hipMemcpyDeviceToDevice,
+          stream(Channel::Coll)));
+    }
+    uint64_t id = make_gpu_request(Channel::Coll, at::Tensor());
+    // keep every buffer alive until the wait
+    {
+      auto& tab = requests();
+      std::lock_guard<std::mutex> tg(tab.mu);
+      if (Request* rq = tab.peek(id)) rq->buffers = all;
+    }
+    return id;
   }
   uint64_t iallgather(const at::Tensor& in, at::Tensor& out) override {
     std::lock_guard<std::mutex> g(mu_);
@@ -1032,6 +1146,7 @@ void wait_request(uint64_t id) {
   } else if (r.work) {
     wait_work(r.work);
   }
+  for (auto& w : r.works) wait_work(w);
   // Send side checks after the data wait (data delivered => the peer posted
   // its recv => its meta recv is posted too; no added blocking).
   if (r.debug_expect.empty()) check_debug_handshake(r);

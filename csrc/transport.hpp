@@ -72,6 +72,14 @@ struct Transport : std::enable_shared_from_this<Transport> {
                         const std::vector<int>& speers,
                         std::vector<at::Tensor>& recvbufs,
                         const std::vector<int>& rpeers) = 0;
+  // Non-blocking exchange: same grouped launch, but the caller's stream is
+  // only joined when the returned request is waited. One phase of the
+  // pipelined (chunked) axis collectives — phase k+1's pack and phase k's
+  // unpack overlap phase k's wire time.
+  virtual uint64_t iexchange(const std::vector<at::Tensor>& sendbufs,
+                             const std::vector<int>& speers,
+                             std::vector<at::Tensor>& recvbufs,
+                             const std::vector<int>& rpeers) = 0;
   // Non-blocking p2p. Returns a request id resolvable via wait_request().
   // Matching contract (identical on ALL transports): FIFO per
   // (peer, channel) — the n-th send posted to a peer on a channel matches

@@ -268,6 +268,29 @@ def test_hierarchical_fp8_bitwise_full_path(world1):
         m._C.force_full_path(False)
 
 
+def test_phased_pipeline_gpu(world1):
+    m = world1
+    comm = m.COMM_WORLD
+    # force K=4 phases and the full path at world 1: phased slab pack ->
+    # grouped iexchange (self) -> phased unpack, on the CDNA4 kernels
+    os.environ["MPI4TORCH_AMD_PIPELINE_MB"] = "0.0001"
+    m._C.reload_config()
+    m._C.force_full_path(True)
+    try:
+        x = torch.rand(6, 8, 4, device="cuda", dtype=torch.bfloat16)
+        y = comm.Alltoall(x, 1, 0, 6)
+        torch.testing.assert_close(y.float(), x.float())
+        z = torch.rand(3, 10, 2, device="cuda").requires_grad_()
+        w = comm.Alltoall(z, 2, 1, 10)
+        torch.testing.assert_close(w, z.detach())
+        w.sum().backward()
+        torch.testing.assert_close(z.grad, torch.ones_like(z))
+    finally:
+        os.environ["MPI4TORCH_AMD_PIPELINE_MB"] = "64"
+        m._C.reload_config()
+        m._C.force_full_path(False)
+
+
 def test_hipgraph_capture(world1):
     m = world1
     comm = m.COMM_WORLD

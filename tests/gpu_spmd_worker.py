@@ -229,6 +229,39 @@ def section_fp8_bitwise_pairloc(m, comm, device):
         assert gmin[i, 0] == mn and gmin[i, 1] == wl, (i, gmin[i])
 
 
+def section_phased_pipeline(m, comm, device):
+    # chunked pack->wire pipelining at world > 1: force K=4 phases and
+    # check values + identities + adjoints against closed forms
+    rank, world = comm.rank, comm.size
+    os.environ["MPI4TORCH_AMD_PIPELINE_MB"] = "0.0001"
+    m._C.reload_config()
+    try:
+        # recv side marshaled (EP/bench layout)
+        x = (torch.arange(world * 3 * 7 * 2, dtype=torch.float32,
+                          device=device).reshape(world * 3, 7, 2)
+             * (rank + 1)).requires_grad_()
+        y = comm.Alltoall(x, 1, 0, 3)
+        base = torch.arange(world * 3 * 7 * 2, dtype=torch.float32,
+                            device=device).reshape(world * 3, 7, 2)
+        for r in range(world):
+            want = base[rank * 3:(rank + 1) * 3] * (r + 1)
+            assert (y.detach()[:, r * 7:(r + 1) * 7] == want).all()
+        z = comm.Alltoall(y, 0, 1, 7)
+        assert (z.detach() == x.detach()).all()
+        z.sum().backward()
+        assert (x.grad == 1.0).all()
+        # send side marshaled + variable sizes
+        a = torch.randn(rank + 2, world * 4, 3, device=device).requires_grad_()
+        b = comm.Alltoall(a, 0, 1, 4)
+        c = comm.Alltoall(b, 1, 0, rank + 2)
+        assert torch.allclose(c.detach(), a.detach())
+        c.sum().backward()
+        assert (a.grad == 1.0).all()
+    finally:
+        os.environ["MPI4TORCH_AMD_PIPELINE_MB"] = "64"
+        m._C.reload_config()
+
+
 def section_reducescatter(m, comm, device):
     rank, world = comm.rank, comm.size
     x = torch.full((world * 4, 3), 1.0, device=device).requires_grad_()
@@ -336,6 +369,8 @@ def main():
     section_iallreduce_overlap(m, comm, device)
     log(rank, "fp8 / bitwise / pairloc")
     section_fp8_bitwise_pairloc(m, comm, device)
+    log(rank, "phased pipelining (forced K=4)")
+    section_phased_pipeline(m, comm, device)
     log(rank, "reducescatter")
     section_reducescatter(m, comm, device)
     log(rank, "alltoall pairwise (EP)")

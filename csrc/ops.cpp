@@ -1431,8 +1431,15 @@ Tensor Communicator::AlltoallvImpl(const Tensor& input, int64_t gatheraxis,
     // at identical byte offsets. Supported: exactly one marshaled side
     // (different axes), or both (same axis — rows coincide); the rare
     // both-marshaled different-axis case falls back to one phase.
+    bool capturing = false;
+    if (in.is_cuda()) {
+      hipStreamCaptureStatus st = hipStreamCaptureStatusNone;
+      (void)hipStreamIsCapturing(current_gpu_stream(in), &st);
+      capturing = st != hipStreamCaptureStatusNone;
+    }
     int64_t K = 1;
-    if (config().pipeline_chunk_bytes > 0 && (send_packed || recv_packed) &&
+    if (!capturing && config().pipeline_chunk_bytes > 0 &&
+        (send_packed || recv_packed) &&
         (send_axis == recv_axis || send_packed != recv_packed)) {
       int64_t global_elems = 1;
       for (size_t d = 0; d < outsizes.size(); ++d) {

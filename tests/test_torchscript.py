@@ -60,6 +60,57 @@ def _scripted_axis_worker(rank, world):
     assert out.shape[0] == 2 * world and out.shape[1] == 3 * world
 
 
+def _scripted_extensions_worker(rank, world):
+    comm = m.COMM_WORLD
+
+    @torch.jit.script
+    def ep_roundtrip(x: torch.Tensor, comm_: m.MPI_Communicator,
+                     world: int) -> torch.Tensor:
+        counts = [2 for _ in range(world)]
+        sizes = [5 for _ in range(world)]
+        y = comm_.Alltoallv(x, 1, 0, counts, sizes)
+        return comm_.Alltoallv(y, 0, 1, sizes, counts)
+
+    @torch.jit.script
+    def pairwise(x: torch.Tensor, comm_: m.MPI_Communicator,
+                 world: int) -> torch.Tensor:
+        sc = [1 for _ in range(world)]
+        empty: list[int] = []
+        return comm_.AlltoallPairwise(x, 0, sc, empty)
+
+    @torch.jit.script
+    def overlap(x: torch.Tensor, comm_: m.MPI_Communicator) -> torch.Tensor:
+        h = comm_.Iallgather(x)
+        return comm_.Wait(h)
+
+    @torch.jit.script
+    def minloc(x: torch.Tensor, comm_: m.MPI_Communicator) -> torch.Tensor:
+        return comm_.Allreduce(x, m.MPI_MINLOC)
+
+    x = torch.rand(world * 2, 5, dtype=torch.double).requires_grad_()
+    rt = ep_roundtrip(x, comm, world)
+    assert (rt.detach() == x.detach()).all()
+    rt.sum().backward()
+    assert (x.grad == 1.0).all()
+
+    p = torch.full((world, 3), float(rank), dtype=torch.double)
+    got = pairwise(p, comm, world)
+    for r in range(world):
+        assert (got[r] == r).all()
+
+    g = overlap(torch.full((4,), float(rank + 1)), comm)
+    assert g.numel() == 4 * world
+
+    pairs = torch.stack([torch.full((3,), float(rank)),
+                         torch.full((3,), float(rank * 7))], dim=-1)
+    mn = minloc(pairs, comm)
+    assert (mn[:, 0] == 0).all() and (mn[:, 1] == 0).all()
+
+
+def test_scripted_extensions_ws3():
+    run_spmd(3, _scripted_extensions_worker)
+
+
 def test_scripted_allreduce_ws2():
     run_spmd(2, _scripted_allreduce_worker)
 

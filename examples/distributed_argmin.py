@@ -11,9 +11,13 @@ Run: torchrun --standalone --local-addr 127.0.0.1 --nproc-per-node 4 \
         examples/distributed_argmin.py
 """
 
+import os
+import sys
+
 import torch
 import torch.distributed as dist
 
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import mpi4torch_amd as m4a
 
 comm = m4a.COMM_WORLD

@@ -81,6 +81,11 @@ struct Config {
   // indefinitely (default). (GPU-side progress is stream-ordered; use the
   // desync detector for pre-enqueue divergence.)
   int64_t op_timeout_ms = 0;
+  // MPI4TORCH_AMD_FORCE_HIERARCHICAL=1 (testing): run the hierarchical
+  // allreduce lowering (bitwise/fp8/pairloc) on the CPU transport too,
+  // with torch-composite local reductions — lets the gloo SPMD suite
+  // validate the multi-rank block-exchange geometry the GPU uses.
+  bool force_hierarchical = false;
   // MPI4TORCH_AMD_PIPELINE_MB (float, default 64): chunk size for the
   // phased pack->wire pipelining of axis-marshaling collectives. Payloads
   // whose packed side exceeds one chunk are exchanged in up to 4 phases so

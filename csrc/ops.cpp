@@ -418,10 +418,12 @@ Tensor hierarchical_allreduce(Transport& tr, const Tensor& in, int64_t unit,
   // phase 2: fused local reduction of the P copies of the owned block
   auto myblock = nz_narrow(out_flat, displs[me], counts[me]);
   if (counts[me] > 0) local_reduce(staging, myblock, counts[me], P);
-  // phase 3: allgather the reduced blocks. Equal counts: one RCCL
+  // phase 3: allgather the reduced blocks. Equal counts on GPU: one RCCL
   // allgather, in place (myblock aliases out_flat at the rank-major
-  // offset — the NCCL in-place convention). Unequal: grouped p2p.
-  if (nunits % P == 0) {
+  // offset — the NCCL in-place convention). Unequal counts, or the CPU
+  // test mode (gloo's allgather does not define input/output aliasing):
+  // grouped p2p.
+  if (nunits % P == 0 && tr.is_gpu()) {
     tr.allgather_equal(myblock, out_flat);
   } else {
     std::vector<Tensor> s2, r2;
@@ -482,12 +484,18 @@ Tensor pairloc_lowered(Transport& tr, const Tensor& in, int64_t op) {
   const int kop = op == kMinLoc ? 0 : 1;
   const int P = tr.size();
   const int dt = pairloc_dtype_code(in.scalar_type());  // dtype validation
-  if (tr.is_gpu()) {
+  if (tr.is_gpu() || config().force_hierarchical) {
     return hierarchical_allreduce(
         tr, in, /*unit=*/2,
         [&](const Tensor& stg, Tensor& blk, int64_t n, int nr) {
-          launch_pairloc_reduce(stg.data_ptr(), blk.data_ptr(), n / 2, nr,
-                                kop, dt, current_gpu_stream(in));
+          if (blk.is_cuda()) {
+            launch_pairloc_reduce(stg.data_ptr(), blk.data_ptr(), n / 2,
+                                  nr, kop, dt, current_gpu_stream(in));
+          } else {
+            blk.copy_(pairloc_local_reduce(stg.view({nr, -1, 2}), kop,
+                                           in.scalar_type())
+                          .view({-1}));
+          }
         });
   }
   // CPU: allgather + torch composite arg-reduce
@@ -531,7 +539,7 @@ Tensor allreduce_lowered(Transport& tr, const Tensor& in, int64_t op) {
     TORCH_CHECK(at::isIntegralType(in.scalar_type(), /*includeBool=*/true),
                 "mpi4torch_amd: ", red_op_name(op),
                 " requires an integral tensor");
-    if (!tr.is_gpu()) {
+    if (!tr.is_gpu() && !config().force_hierarchical) {
       auto out = at::empty_like(in);
       tr.allreduce(in, out, (RedOp)op);
       return out;
@@ -543,8 +551,19 @@ Tensor allreduce_lowered(Transport& tr, const Tensor& in, int64_t op) {
     return hierarchical_allreduce(
         tr, in, /*unit=*/1,
         [&](const Tensor& stg, Tensor& blk, int64_t n, int nr) {
-          launch_bitwise_reduce(stg.data_ptr(), blk.data_ptr(), n * esize,
-                                nr, kop, current_gpu_stream(in));
+          if (blk.is_cuda()) {
+            launch_bitwise_reduce(stg.data_ptr(), blk.data_ptr(), n * esize,
+                                  nr, kop, current_gpu_stream(in));
+            return;
+          }
+          auto v = stg.view({nr, n});
+          auto acc = v[0].clone();
+          for (int64_t r = 1; r < nr; ++r) {
+            if (kop == 0) acc.bitwise_and_(v[r]);
+            else if (kop == 1) acc.bitwise_or_(v[r]);
+            else acc.bitwise_xor_(v[r]);
+          }
+          blk.copy_(acc);
         });
   }
   TORCH_CHECK(is_arith(op));
@@ -555,7 +574,7 @@ Tensor allreduce_lowered(Transport& tr, const Tensor& in, int64_t op) {
   }
   const bool fp8 = in.scalar_type() == at::kFloat8_e4m3fn ||
                    in.scalar_type() == at::kFloat8_e5m2;
-  if (fp8 && tr.is_gpu()) {
+  if (fp8 && (tr.is_gpu() || config().force_hierarchical)) {
     // fp8 stays fp8 on the wire: hierarchical block exchange + one fused
     // CDNA4 reduction per element with fp32 accumulators and a single
     // quantization (better numerics than a per-hop-quantizing ring, 4x
@@ -566,8 +585,20 @@ Tensor allreduce_lowered(Transport& tr, const Tensor& in, int64_t op) {
     return hierarchical_allreduce(
         tr, in, /*unit=*/1,
         [&](const Tensor& stg, Tensor& blk, int64_t n, int nr) {
-          launch_fp8_reduce(stg.data_ptr(), blk.data_ptr(), n, nr, kop,
-                            e5m2, current_gpu_stream(in));
+          if (blk.is_cuda()) {
+            launch_fp8_reduce(stg.data_ptr(), blk.data_ptr(), n, nr, kop,
+                              e5m2, current_gpu_stream(in));
+            return;
+          }
+          auto v = stg.view({nr, n}).to(at::kFloat);
+          auto acc = v[0];
+          for (int64_t r = 1; r < nr; ++r) {
+            if (kop == 0) acc = acc + v[r];
+            else if (kop == 1) acc = acc * v[r];
+            else if (kop == 2) acc = at::minimum(acc, v[r]);
+            else acc = at::maximum(acc, v[r]);
+          }
+          blk.copy_(acc.to(in.scalar_type()));
         });
   }
   auto up = in.to(upcast_for_reduce(in.scalar_type()));

@@ -38,6 +38,8 @@ static void config_from_env(Config& c) {
   if (const char* pc = std::getenv("MPI4TORCH_AMD_PIPELINE_MB")) {
     c.pipeline_chunk_bytes = (int64_t)(std::atof(pc) * 1048576.0);
   }
+  const char* fh = std::getenv("MPI4TORCH_AMD_FORCE_HIERARCHICAL");
+  c.force_hierarchical = fh && fh[0] == '1';
 }
 
 Config& config() {

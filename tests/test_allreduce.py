@@ -199,6 +199,70 @@ def _pairloc_worker(rank, world):
         assert (ri[1:, 1] == world - 1).all()
 
 
+def _hierarchical_worker(rank, world):
+    # The GPU lowering for ops RCCL cannot reduce on the wire (bitwise,
+    # fp8, minloc/maxloc) is a hierarchical block exchange + local reduce
+    # + allgather. MPI4TORCH_AMD_FORCE_HIERARCHICAL=1 runs the SAME
+    # multi-rank exchange geometry on gloo with torch local reductions, so
+    # phases 1-3 (incl. non-divisible block sizes) are CI-validated at
+    # world > 1 before any multi-GPU run.
+    import os
+
+    os.environ["MPI4TORCH_AMD_FORCE_HIERARCHICAL"] = "1"
+    import mpi4torch_amd as m
+
+    comm = m.COMM_WORLD
+    for n in (4 * world, 4 * world + 3, 1, world - 1 if world > 1 else 1):
+        # bitwise vs gloo-native reference
+        ti = (torch.arange(n, dtype=torch.int64) * (rank + 7)) % (1 << 20)
+        m._C.reload_config()
+        for op in (m.MPI_BAND, m.MPI_BOR, m.MPI_BXOR):
+            got = comm.Allreduce(ti, op)
+            os.environ["MPI4TORCH_AMD_FORCE_HIERARCHICAL"] = "0"
+            m._C.reload_config()
+            want = comm.Allreduce(ti, op)  # gloo-native path
+            os.environ["MPI4TORCH_AMD_FORCE_HIERARCHICAL"] = "1"
+            m._C.reload_config()
+            assert (got == want).all(), (n, op)
+
+        # fp8 vs the upcast reference
+        t8 = (torch.randn(n) * 0.25 * (rank + 1)).to(torch.float8_e4m3fn)
+        got8 = comm.Allreduce(t8, m.MPI_SUM)
+        allf = comm.Allgather(t8.float().reshape(1, n), 0)
+        ref = allf[0]
+        for r in range(1, world):
+            ref = ref + allf[r]
+        want8 = ref.to(torch.float8_e4m3fn)
+        assert (got8.view(torch.uint8) == want8.view(torch.uint8)).all(), n
+
+    # pairloc through the hierarchical path (unit=2 blocks)
+    for npairs in (2 * world + 1, 3):
+        vals = torch.tensor([(i + rank) % world for i in range(npairs)],
+                            dtype=torch.float64)
+        locs = torch.full((npairs,), float(rank * 10), dtype=torch.float64)
+        pairs = torch.stack([vals, locs], dim=-1)
+        got = comm.Allreduce(pairs, m.MPI_MINLOC)
+        av = comm.Allgather(vals.reshape(1, -1), 0)
+        al = comm.Allgather(locs.reshape(1, -1), 0)
+        for i in range(npairs):
+            col = av[:, i]
+            mn = col.min()
+            wl = al[:, i][col == mn].min()
+            assert got[i, 0] == mn and got[i, 1] == wl, (npairs, i)
+
+
+def test_hierarchical_ws2():
+    run_spmd(2, _hierarchical_worker)
+
+
+def test_hierarchical_ws5():
+    run_spmd(5, _hierarchical_worker)
+
+
+def test_hierarchical_ws7():
+    run_spmd(7, _hierarchical_worker)
+
+
 def test_allreduce_pairloc_ws2():
     run_spmd(2, _pairloc_worker)
 

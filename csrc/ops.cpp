@@ -35,6 +35,7 @@
 #include <algorithm>
 #include <cstdlib>
 #include <cstring>
+#include <functional>
 #include <limits>
 #include <map>
 #include <mutex>
@@ -223,6 +224,59 @@ std::vector<int64_t> prefix_displs(const std::vector<int64_t>& counts) {
   std::vector<int64_t> d(counts.size(), 0);
   for (size_t i = 1; i < counts.size(); ++i) d[i] = d[i - 1] + counts[i - 1];
   return d;
+}
+
+// Phase count for the chunked pack->wire pipelining. MUST be computed
+// from rank-shared quantities only (global logical payload), so every
+// rank slices every pair block at identical offsets. Graph capture
+// forces one phase (iexchange+wait cannot be captured).
+int64_t phase_count(const Tensor& t, int64_t per_rank_bytes) {
+  const int64_t c = config().pipeline_chunk_bytes;
+  if (c <= 0) return 1;
+  if (t.is_cuda()) {
+    hipStreamCaptureStatus st = hipStreamCaptureStatusNone;
+    (void)hipStreamIsCapturing(current_gpu_stream(t), &st);
+    if (st != hipStreamCaptureStatusNone) return 1;
+  }
+  return std::min<int64_t>(
+      4, std::max<int64_t>(1, (per_rank_bytes + c - 1) / c));
+}
+
+// Phased exchange for the funnel collectives (Gather/Scatter/Allgather):
+// every pair block shares the SAME row count (the common `before` of the
+// marshaled layout); per-pair row lengths come in elements. pack_phase /
+// unpack_phase (either may be null) marshal one [row_lo, row_hi) slice;
+// each phase is one full-width grouped iexchange, so marshaling overlaps
+// the previous phase's wire time.
+void phased_funnel_exchange(
+    Transport& tr, int64_t K, int64_t rows,
+    const std::vector<Tensor>& sblocks, const std::vector<int>& speers,
+    const std::vector<int64_t>& srowe, std::vector<Tensor>& rblocks,
+    const std::vector<int>& rpeers, const std::vector<int64_t>& rrowe,
+    const std::function<void(int64_t, int64_t)>& pack_phase,
+    const std::function<void(int64_t, int64_t)>& unpack_phase) {
+  auto bound = [&](int64_t k) { return rows * k / K; };
+  auto slice = [](const Tensor& blk, int64_t lo, int64_t len) {
+    auto flat = blk.view({-1});
+    return len > 0 ? flat.narrow(0, lo, len) : flat.narrow(0, 0, 0);
+  };
+  std::vector<uint64_t> reqs((size_t)K);
+  for (int64_t k = 0; k < K; ++k) {
+    const int64_t b0 = bound(k), b1 = bound(k + 1);
+    if (pack_phase) pack_phase(b0, b1);
+    std::vector<Tensor> sk(sblocks.size()), rk(rblocks.size());
+    for (size_t i = 0; i < sblocks.size(); ++i) {
+      sk[i] = slice(sblocks[i], b0 * srowe[i], (b1 - b0) * srowe[i]);
+    }
+    for (size_t j = 0; j < rblocks.size(); ++j) {
+      rk[j] = slice(rblocks[j], b0 * rrowe[j], (b1 - b0) * rrowe[j]);
+    }
+    reqs[(size_t)k] = tr.iexchange(sk, speers, rk, rpeers);
+  }
+  for (int64_t k = 0; k < K; ++k) {
+    wait_request(reqs[(size_t)k]);
+    if (unpack_phase) unpack_phase(bound(k), bound(k + 1));
+  }
 }
 
 // ------------------------- reduction lowering -----------------------------
@@ -972,20 +1026,49 @@ Tensor Communicator::Gather(const Tensor& input, int64_t gatheraxis,
     newsizes[gatheraxis] = (me == (int)root) ? total : 0;
     auto out = at::empty(newsizes, in.options());
 
+    // phased pipelining: rows = the common `before` (non-axis dims match
+    // across ranks); each sender's contiguous slab rows land directly in
+    // root's per-peer staging, unpacked phase by phase
+    const int64_t K =
+        g.before > 1
+            ? phase_count(in, g.before * total * g.after *
+                                  in.element_size() / tr.size())
+            : 1;
+
     if (me == (int)root) {
       auto blocks = make_blocks(in, g.before, g.after, counts, displs, &out);
       std::vector<Tensor> sends{in};
       std::vector<int> speers{(int)root};
       auto rpeers = iota_peers(tr.size());
-      tr.exchange(sends, speers, blocks, rpeers);
-      if (g.before != 1) {
-        move_axis_blocks(out, gatheraxis, displs, counts, blocks,
-                         /*pack=*/false);
+      if (K > 1) {
+        std::vector<int64_t> srowe{g.axis * g.after};
+        std::vector<int64_t> rrowe(counts.size());
+        for (size_t i = 0; i < counts.size(); ++i) {
+          rrowe[i] = counts[i] * g.after;
+        }
+        phased_funnel_exchange(
+            tr, K, g.before, sends, speers, srowe, blocks, rpeers, rrowe,
+            nullptr, [&](int64_t b0, int64_t b1) {
+              move_axis_blocks(out, gatheraxis, displs, counts, blocks,
+                               /*pack=*/false, b0, b1);
+            });
+      } else {
+        tr.exchange(sends, speers, blocks, rpeers);
+        if (g.before != 1) {
+          move_axis_blocks(out, gatheraxis, displs, counts, blocks,
+                           /*pack=*/false);
+        }
       }
     } else {
       std::vector<Tensor> sends{in}, recvs;
       std::vector<int> speers{(int)root}, rpeers;
-      tr.exchange(sends, speers, recvs, rpeers);
+      if (K > 1) {
+        std::vector<int64_t> srowe{g.axis * g.after}, rrowe;
+        phased_funnel_exchange(tr, K, g.before, sends, speers, srowe, recvs,
+                               rpeers, rrowe, nullptr, nullptr);
+      } else {
+        tr.exchange(sends, speers, recvs, rpeers);
+      }
     }
     return stager.from_comm(std::move(out));
   }();
@@ -1038,21 +1121,58 @@ Tensor Communicator::Scatter(const Tensor& input, int64_t scatteraxis,
     outsizes[axis] = counts[me];
     auto out = at::empty(outsizes, in.options());
 
+    // phased pipelining (marshaled side = root's pack; receivers' out IS
+    // the slab layout, so their phase slices are plain contiguous ranges).
+    // `before` comes from the SHARED rootsizes, so non-root ranks compute
+    // the same K and row geometry without touching their placeholder.
+    int64_t before_c = 1, after_c = 1;
+    for (int64_t d = 0; d < ndim; ++d) {
+      if (d < axis) before_c *= rootsizes[d];
+      else if (d > axis) after_c *= rootsizes[d];
+    }
+    const int64_t K =
+        before_c > 1
+            ? phase_count(in, before_c * rootsizes[axis] * after_c *
+                                  in.element_size() / tr.size())
+            : 1;
+
     if (me == (int)root) {
       TORCH_CHECK(in.sizes().vec() == rootsizes);
       const auto g = axis_geom(in, axis);
       auto blocks = make_blocks(in, g.before, g.after, counts, displs, &in);
-      if (g.before != 1) {
-        move_axis_blocks(in, axis, displs, counts, blocks, /*pack=*/true);
-      }
       auto speers = iota_peers(tr.size());
       std::vector<Tensor> recvs{out};
       std::vector<int> rpeers{(int)root};
-      tr.exchange(blocks, speers, recvs, rpeers);
+      if (K > 1) {
+        std::vector<int64_t> srowe(counts.size());
+        for (size_t j = 0; j < counts.size(); ++j) {
+          srowe[j] = counts[j] * g.after;
+        }
+        std::vector<int64_t> rrowe{counts[me] * g.after};
+        phased_funnel_exchange(
+            tr, K, g.before, blocks, speers, srowe, recvs, rpeers, rrowe,
+            [&](int64_t b0, int64_t b1) {
+              move_axis_blocks(in, axis, displs, counts, blocks,
+                               /*pack=*/true, b0, b1);
+            },
+            nullptr);
+      } else {
+        if (g.before != 1) {
+          move_axis_blocks(in, axis, displs, counts, blocks, /*pack=*/true);
+        }
+        tr.exchange(blocks, speers, recvs, rpeers);
+      }
     } else {
       std::vector<Tensor> sends, recvs{out};
       std::vector<int> speers, rpeers{(int)root};
-      tr.exchange(sends, speers, recvs, rpeers);
+      if (K > 1) {
+        std::vector<int64_t> srowe;
+        std::vector<int64_t> rrowe{counts[me] * after_c};
+        phased_funnel_exchange(tr, K, before_c, sends, speers, srowe, recvs,
+                               rpeers, rrowe, nullptr, nullptr);
+      } else {
+        tr.exchange(sends, speers, recvs, rpeers);
+      }
     }
     return stager.from_comm(std::move(out));
   }();
@@ -1113,9 +1233,36 @@ Tensor Communicator::Allgather(const Tensor& input, int64_t gatheraxis) {
     newsizes[gatheraxis] = total;
     auto out = at::empty(newsizes, in.options());
 
+    // phased pipelining: every rank both sends and receives ~the whole
+    // gathered tensor, so the per-rank payload is the global logical size
+    const int64_t K =
+        g.before > 1
+            ? phase_count(in,
+                          g.before * total * g.after * in.element_size())
+            : 1;
     if (equal && g.before == 1) {
       auto out_flat = out.view({-1});
       tr.allgather_equal(in, out_flat);
+    } else if (K > 1) {
+      // grouped p2p with phases: senders' slab rows go straight into
+      // per-peer staging; each phase's unpack overlaps the next phase's
+      // wire (on the xGMI crossbar direct p2p moves the same bytes as
+      // the allgather ring, spread over all links)
+      auto blocks = make_blocks(in, g.before, g.after, counts, displs,
+                                nullptr);
+      std::vector<Tensor> sends((size_t)tr.size(), in);
+      auto peers = iota_peers(tr.size());
+      std::vector<int64_t> srowe((size_t)tr.size(), g.axis * g.after);
+      std::vector<int64_t> rrowe(counts.size());
+      for (size_t i = 0; i < counts.size(); ++i) {
+        rrowe[i] = counts[i] * g.after;
+      }
+      phased_funnel_exchange(
+          tr, K, g.before, sends, peers, srowe, blocks, peers, rrowe,
+          nullptr, [&](int64_t b0, int64_t b1) {
+            move_axis_blocks(out, gatheraxis, displs, counts, blocks,
+                             /*pack=*/false, b0, b1);
+          });
     } else if (equal && tr.is_gpu()) {
       auto staging = at::empty({total * g.before * g.after}, in.options());
       tr.allgather_equal(in, staging);
@@ -1462,15 +1609,8 @@ Tensor Communicator::AlltoallvImpl(const Tensor& input, int64_t gatheraxis,
     // at identical byte offsets. Supported: exactly one marshaled side
     // (different axes), or both (same axis — rows coincide); the rare
     // both-marshaled different-axis case falls back to one phase.
-    bool capturing = false;
-    if (in.is_cuda()) {
-      hipStreamCaptureStatus st = hipStreamCaptureStatusNone;
-      (void)hipStreamIsCapturing(current_gpu_stream(in), &st);
-      capturing = st != hipStreamCaptureStatusNone;
-    }
     int64_t K = 1;
-    if (!capturing && config().pipeline_chunk_bytes > 0 &&
-        (send_packed || recv_packed) &&
+    if ((send_packed || recv_packed) &&
         (send_axis == recv_axis || send_packed != recv_packed)) {
       int64_t global_elems = 1;
       for (size_t d = 0; d < outsizes.size(); ++d) {
@@ -1478,11 +1618,7 @@ Tensor Communicator::AlltoallvImpl(const Tensor& input, int64_t gatheraxis,
                             ? (sdispls.back() + scounts.back())
                             : outsizes[d];
       }
-      const int64_t per_rank_b = global_elems / P * in.element_size();
-      K = std::min<int64_t>(
-          4, std::max<int64_t>(
-                 1, (per_rank_b + config().pipeline_chunk_bytes - 1) /
-                        config().pipeline_chunk_bytes));
+      K = phase_count(in, global_elems / P * in.element_size());
     }
 
     if (K > 1) {

@@ -257,6 +257,20 @@ def section_phased_pipeline(m, comm, device):
         assert torch.allclose(c.detach(), a.detach())
         c.sum().backward()
         assert (a.grad == 1.0).all()
+        # phased funnel collectives (Gather/Scatter/Allgather, before>1)
+        my = rank + 1
+        total = world * (world + 1) // 2
+        fx = torch.full((3, my, 2), float(rank), device=device)
+        fg = comm.Gather(fx, 1, 0)
+        if rank == 0:
+            off = 0
+            for r in range(world):
+                assert (fg[:, off:off + r + 1] == r).all()
+                off += r + 1
+        fb = comm.Scatter(fg, 1, my, 0)
+        assert (fb == fx).all()
+        fa = comm.Allgather(fx, 1)
+        assert fa.shape == (3, total, 2)
     finally:
         os.environ["MPI4TORCH_AMD_PIPELINE_MB"] = "64"
         m._C.reload_config()

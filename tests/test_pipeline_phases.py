@@ -76,6 +76,61 @@ def _phased_alltoall_worker(rank, world):
     assert (v.grad == 1.0).all()
 
 
+def _phased_funnel_worker(rank, world):
+    os.environ["MPI4TORCH_AMD_PIPELINE_MB"] = "0.0001"  # force K=4
+    import mpi4torch_amd as m
+
+    comm = m.COMM_WORLD
+
+    # Gather with before>1 and variable per-rank counts, fwd + adjoint
+    my = rank + 1
+    x = torch.full((3, my, 2), float(rank), dtype=torch.float64)
+    x.requires_grad_()
+    g = comm.Gather(x, 1, 0)
+    total = world * (world + 1) // 2
+    if rank == 0:
+        assert g.shape == (3, total, 2)
+        off = 0
+        for r in range(world):
+            assert (g[:, off:off + r + 1] == r).all()
+            off += r + 1
+    else:
+        assert g.size(1) == 0
+    # Scatter∘Gather identity through the phased paths
+    back = comm.Scatter(g, 1, my, 0)
+    assert (back.detach() == x.detach()).all()
+    back.sum().backward()
+    assert (x.grad == 1.0).all()
+
+    # Allgather with before>1, non-uniform grads (reduce-scatter adjoint)
+    y = torch.full((2, my, 3), float(rank + 1), dtype=torch.float64)
+    y.requires_grad_()
+    ag = comm.Allgather(y, 1)
+    assert ag.shape == (2, total, 3)
+    off = 0
+    for r in range(world):
+        assert (ag.detach()[:, off:off + r + 1] == r + 1).all()
+        off += r + 1
+    ag.backward(torch.full_like(ag, float(rank + 1)))
+    want = sum(r + 1 for r in range(world))
+    assert (y.grad == want).all()
+
+    # equal counts too (the allgather_equal bypass must not kick in when
+    # phased; values must agree regardless)
+    z = torch.full((2, 4, 3), float(rank), dtype=torch.float64)
+    agz = comm.Allgather(z, 1)
+    for r in range(world):
+        assert (agz[:, r * 4:(r + 1) * 4] == r).all()
+
+
+def test_phased_funnel_ws2():
+    run_spmd(2, _phased_funnel_worker)
+
+
+def test_phased_funnel_ws5():
+    run_spmd(5, _phased_funnel_worker)
+
+
 def test_phased_alltoall_ws2():
     run_spmd(2, _phased_alltoall_worker)
 

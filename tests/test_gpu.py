@@ -285,6 +285,15 @@ def test_phased_pipeline_gpu(world1):
         torch.testing.assert_close(w, z.detach())
         w.sum().backward()
         torch.testing.assert_close(z.grad, torch.ones_like(z))
+        # funnel collectives through the phased path (before > 1)
+        f = torch.rand(5, 6, 3, device="cuda", dtype=torch.bfloat16)
+        torch.testing.assert_close(comm.Gather(f, 1, 0).float(), f.float())
+        torch.testing.assert_close(comm.Allgather(f, 1).float(), f.float())
+        torch.testing.assert_close(
+            comm.Scatter(f, 1, 6, 0).float(), f.float())
+        fg = torch.rand(4, 7, 2, device="cuda").requires_grad_()
+        comm.Gather(fg, 1, 0).sum().backward()
+        torch.testing.assert_close(fg.grad, torch.ones_like(fg))
     finally:
         os.environ["MPI4TORCH_AMD_PIPELINE_MB"] = "64"
         m._C.reload_config()

@@ -1609,9 +1609,36 @@ Tensor Communicator::AlltoallvImpl(const Tensor& input, int64_t gatheraxis,
     // at identical byte offsets. Supported: exactly one marshaled side
     // (different axes), or both (same axis — rows coincide); the rare
     // both-marshaled different-axis case falls back to one phase.
+    // SHARED marshaling flags. The LOCAL gs.before/gr.before can differ
+    // across ranks in the different-axis case (the other, rank-sized axis
+    // may lie inside `before`; a rank whose extent there is 1 sees
+    // before==1). The phasing DECISION and the per-pair geometry selection
+    // must be identical on every rank, so they use `before` computed with
+    // the rank-dependent axis at its MAX over the shared size vectors;
+    // the local pack/unpack still use the local flags (a 1-row rank just
+    // contributes empty phases, which both endpoints compute identically).
+    int64_t send_before_shared = 1, recv_before_shared = 1;
+    {
+      const int64_t src_max =
+          *std::max_element(srcsz.begin(), srcsz.end());
+      const int64_t sc_max =
+          *std::max_element(scounts.begin(), scounts.end());
+      auto insz = in.sizes();
+      for (int64_t d = 0; d < send_axis; ++d) {
+        send_before_shared *=
+            (d == recv_axis && send_axis != recv_axis) ? src_max : insz[d];
+      }
+      for (int64_t d = 0; d < recv_axis; ++d) {
+        recv_before_shared *= (d == scatteraxis && send_axis != recv_axis)
+                                  ? sc_max
+                                  : outsizes[d];
+      }
+    }
+    const bool send_m = send_before_shared > 1;
+    const bool recv_m = recv_before_shared > 1;
     int64_t K = 1;
-    if ((send_packed || recv_packed) &&
-        (send_axis == recv_axis || send_packed != recv_packed)) {
+    if ((send_m || recv_m) &&
+        (send_axis == recv_axis || send_m != recv_m)) {
       int64_t global_elems = 1;
       for (size_t d = 0; d < outsizes.size(); ++d) {
         global_elems *= (d == (size_t)scatteraxis)
@@ -1623,9 +1650,12 @@ Tensor Communicator::AlltoallvImpl(const Tensor& input, int64_t gatheraxis,
 
     if (K > 1) {
       const int me_ = me;
-      // per-pair row geometry of the marshaled side, from shared data
+      // per-pair row geometry of the marshaled side, from shared data.
+      // Branch selection uses the SHARED flags (send_m/recv_m), never the
+      // local before values — a rank whose local before is 1 must still
+      // speak the globally-agreed phase geometry.
       auto send_geom = [&](int j) -> std::pair<int64_t, int64_t> {
-        if (send_axis == recv_axis || send_packed) {
+        if (send_axis == recv_axis || send_m) {
           return {gs.before, s_lens[j] * gs.after};
         }
         // recv side is the marshaled one: receiver j's rows (its out dims
@@ -1640,7 +1670,7 @@ Tensor Communicator::AlltoallvImpl(const Tensor& input, int64_t gatheraxis,
         return {before, srcsz[me_] * after};
       };
       auto recv_geom = [&](int i) -> std::pair<int64_t, int64_t> {
-        if (send_axis == recv_axis || recv_packed) {
+        if (send_axis == recv_axis || recv_m) {
           return {gr.before, r_lens[i] * gr.after};
         }
         // send side is the marshaled one: sender i's rows (its in dims =

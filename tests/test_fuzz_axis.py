@@ -97,9 +97,26 @@ def _fuzz_worker(rank, world, n_iters):
             assert torch.equal(out, expect2), f"alltoall it={it}"
 
 
+def _fuzz_phased_worker(rank, world, n_iters):
+    # identical sweep through the phased (chunked) pipeline: 100-byte
+    # chunks force K=4 phases on every geometry that marshals
+    import os
+
+    os.environ["MPI4TORCH_AMD_PIPELINE_MB"] = "0.0001"
+    _fuzz_worker(rank, world, n_iters)
+
+
 def test_fuzz_ws2():
     run_spmd(2, _fuzz_worker, 25)
 
 
 def test_fuzz_ws3():
     run_spmd(3, _fuzz_worker, 25)
+
+
+def test_fuzz_phased_ws3():
+    run_spmd(3, _fuzz_phased_worker, 25)
+
+
+def test_fuzz_phased_ws5():
+    run_spmd(5, _fuzz_phased_worker, 15)

@@ -250,10 +250,12 @@ def section_phased_pipeline(m, comm, device):
         assert (z.detach() == x.detach()).all()
         z.sum().backward()
         assert (x.grad == 1.0).all()
-        # send side marshaled + variable sizes
-        a = torch.randn(rank + 2, world * 4, 3, device=device).requires_grad_()
+        # send side marshaled + variable sizes; rank 0 has extent 1, so its
+        # LOCAL slab before is 1 while peers marshal — the shared-flag
+        # phasing decision (not the local one) must drive every rank
+        a = torch.randn(rank + 1, world * 4, 3, device=device).requires_grad_()
         b = comm.Alltoall(a, 0, 1, 4)
-        c = comm.Alltoall(b, 1, 0, rank + 2)
+        c = comm.Alltoall(b, 1, 0, rank + 1)
         assert torch.allclose(c.detach(), a.detach())
         c.sum().backward()
         assert (a.grad == 1.0).all()

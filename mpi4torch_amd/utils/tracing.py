@@ -119,9 +119,38 @@ class TracedCommunicator:
             lambda: self._comm.Alltoall(tensor, gatheraxis, scatteraxis,
                                         numelem))
 
+    def Reducescatter(self, tensor, axis, numelem):
+        return self._record(
+            "Reducescatter", tensor,
+            lambda: self._comm.Reducescatter(tensor, axis, numelem))
+
+    def Alltoallv(self, tensor, gatheraxis, scatteraxis, target_counts,
+                  source_sizes):
+        return self._record(
+            "Alltoallv", tensor,
+            lambda: self._comm.Alltoallv(tensor, gatheraxis, scatteraxis,
+                                         target_counts, source_sizes))
+
+    def AlltoallPairwise(self, tensor, axis, send_counts, recv_counts):
+        return self._record(
+            "AlltoallPairwise", tensor,
+            lambda: self._comm.AlltoallPairwise(tensor, axis, send_counts,
+                                                recv_counts))
+
     def Iallreduce(self, tensor, op):
         return self._record("Iallreduce", tensor,
                             lambda: self._comm.Iallreduce(tensor, op))
+
+    def Ireducescatter(self, tensor, op):
+        return self._record("Ireducescatter", tensor,
+                            lambda: self._comm.Ireducescatter(tensor, op))
+
+    def Iallgather(self, tensor):
+        return self._record("Iallgather", tensor,
+                            lambda: self._comm.Iallgather(tensor))
+
+    def Barrier(self):
+        return self._comm.Barrier()
 
     def Isend(self, tensor, dest, tag):
         return self._record("Isend", tensor,

@@ -21,6 +21,16 @@ def _trace_worker(rank, world):
     comm.clear_trace()
     assert comm.trace_records() == []
 
+    # round-2 surface is traced too
+    comm.Reducescatter(torch.ones(world * 2), 0, 2)
+    comm.Alltoallv(torch.ones(world, 3), 1, 0, [1] * world, [3] * world)
+    comm.AlltoallPairwise(torch.ones(world, 2), 0, [1] * world, [])
+    h = comm.Iallgather(torch.ones(4))
+    comm.Wait(h)
+    ops = [r.op for r in comm.trace_records()]
+    assert ops == ["Reducescatter", "Alltoallv", "AlltoallPairwise",
+                   "Iallgather", "Wait"], ops
+
 
 def test_trace_ws2():
     run_spmd(2, _trace_worker)

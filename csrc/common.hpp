@@ -86,12 +86,16 @@ struct Config {
   // with torch-composite local reductions — lets the gloo SPMD suite
   // validate the multi-rank block-exchange geometry the GPU uses.
   bool force_hierarchical = false;
-  // MPI4TORCH_AMD_PIPELINE_MB (float, default 64): chunk size for the
+  // MPI4TORCH_AMD_PIPELINE_MB (float, default 256): chunk size for the
   // phased pack->wire pipelining of axis-marshaling collectives. Payloads
   // whose packed side exceeds one chunk are exchanged in up to 4 phases so
   // the CDNA4 pack/unpack kernels overlap the wire time of neighboring
-  // phases. <=0 disables phasing.
-  int64_t pipeline_chunk_bytes = 64ll << 20;
+  // phases. <=0 disables phasing. Default sized from measurement
+  // (profiles/tune_pipeline_r2.log): each phase costs ~40 us of
+  // event-chain latency and loses some copy occupancy, while the overlap
+  // can hide ~2*payload/5TB/s of marshaling — net-positive only when the
+  // wire is the slow side (multi-GPU) and payloads are >= ~1 GiB/rank.
+  int64_t pipeline_chunk_bytes = 256ll << 20;
 };
 
 Config& config();

@@ -131,6 +131,10 @@ def test_phased_funnel_ws5():
     run_spmd(5, _phased_funnel_worker)
 
 
+def test_phased_funnel_ws7():
+    run_spmd(7, _phased_funnel_worker)
+
+
 def test_phased_alltoall_ws2():
     run_spmd(2, _phased_alltoall_worker)
 

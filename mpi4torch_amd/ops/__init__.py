@@ -68,6 +68,16 @@ def iallreduce(tensor: torch.Tensor, op: Optional[int] = None, comm=None):
     return _comm(comm).Iallreduce(tensor, _m.MPI_SUM if op is None else op)
 
 
+def ireducescatter(tensor: torch.Tensor, op: Optional[int] = None,
+                   comm=None):
+    return _comm(comm).Ireducescatter(
+        tensor, _m.MPI_SUM if op is None else op)
+
+
+def iallgather(tensor: torch.Tensor, comm=None):
+    return _comm(comm).Iallgather(tensor)
+
+
 def isend(tensor: torch.Tensor, dest: int, tag: int = 0, comm=None):
     return _comm(comm).Isend(tensor, dest, tag)
 

@@ -155,6 +155,23 @@ class FullyShardedDataParallel(torch.nn.Module):
             if any(p.requires_grad for p in m.parameters())
         ]
         assert unit_modules, "no FSDP units found"
+        # Parameters shared ACROSS units (weight tying between units) would
+        # be flattened into two shards and silently diverge; detect and
+        # refuse. Tying WITHIN one unit is fine: nn.Module.parameters()
+        # deduplicates, so the unit flattens one copy that both uses view.
+        seen = {}
+        for m in unit_modules:
+            for name, p in m.named_parameters():
+                if not p.requires_grad:
+                    continue
+                if p in seen:
+                    raise RuntimeError(
+                        "mpi4torch_amd FSDP: parameter is shared between "
+                        f"units ('{seen[p]}' and '{name}'). Tied parameters "
+                        "must live in the SAME unit — pass units=[...] "
+                        "grouping the tied modules together."
+                    )
+                seen[p] = name
         self._units = [_Unit(m, self.comm, master_dtype)
                        for m in unit_modules]
         self._by_module = {u.module: u for u in self._units}

@@ -511,6 +511,69 @@ def test_fsdp_accumulation_ws2():
     run_spmd(2, _fsdp_accum_worker)
 
 
+def _fsdp_tied_worker(rank, world):
+    import mpi4torch_amd as m  # noqa: F401
+    from mpi4torch_amd.parallel import FullyShardedDataParallel
+
+    class Tied(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.embed = torch.nn.Embedding(16, 8)
+            self.mid = torch.nn.Linear(8, 8)
+            self.head = torch.nn.Linear(8, 16, bias=False)
+            self.head.weight = self.embed.weight  # weight tying
+
+        def forward(self, x):
+            return self.head(self.mid(self.embed(x)))
+
+    net = Tied()
+    # tied params in DIFFERENT units: two shards would silently diverge —
+    # must refuse with guidance
+    try:
+        FullyShardedDataParallel(net, units=[net.embed, net.mid, net.head])
+        raise AssertionError("expected shared-parameter detection")
+    except RuntimeError as e:
+        assert "SAME unit" in str(e), e
+
+    # tying WITHIN one invoked unit works: nn.Module.parameters()
+    # deduplicates, so the unit flattens ONE copy both uses view
+    class SelfTied(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.proj = torch.nn.Linear(8, 8, bias=False)
+            self.out = torch.nn.Linear(8, 8, bias=False)
+            self.out.weight = self.proj.weight  # tied inside the unit
+
+        def forward(self, x):
+            return self.out(torch.tanh(self.proj(x)))
+
+    class Net2(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.a = SelfTied()
+            self.b = torch.nn.Linear(8, 4)
+
+        def forward(self, x):
+            return self.b(self.a(x))
+
+    net2 = Net2()
+    model = FullyShardedDataParallel(net2)
+    assert len(model._units[0].params) == 1  # deduplicated tied weight
+    x = torch.randn(4, 8)
+    loss = model(x).sum()
+    loss.backward()
+    model.finish_backward()
+    assert all(u.shard.grad is not None for u in model._units)
+    # tying held after a materialize cycle
+    model._units[0].materialize()
+    assert (net2.a.out.weight.data_ptr()
+            == net2.a.proj.weight.data_ptr())
+
+
+def test_fsdp_tied_params_ws2():
+    run_spmd(2, _fsdp_tied_worker)
+
+
 def _iallgather_worker(rank, world):
     import mpi4torch_amd as m
     from mpi4torch_amd.parallel import FullyShardedDataParallel  # noqa: F401

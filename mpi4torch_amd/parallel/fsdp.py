@@ -136,11 +136,16 @@ class _Unit:
 class FullyShardedDataParallel(torch.nn.Module):
     def __init__(self, module: torch.nn.Module,
                  units: Optional[Sequence[torch.nn.Module]] = None,
-                 comm=None, average: bool = True, master_dtype=None):
+                 comm=None, average: bool = True, master_dtype=None,
+                 prefetch: int = 1):
         super().__init__()
         self.module = module
         self.comm = comm if comm is not None else m4a.COMM_WORLD
         self.average = average
+        # how many upcoming units to allgather ahead of use (0 disables;
+        # each in-flight prefetch holds one unit's full buffer, so the
+        # memory high-water mark is (1 + prefetch) units)
+        self.prefetch = max(0, int(prefetch))
         self._sync_enabled = True
         if self.comm.size > 1:
             with torch.no_grad():
@@ -191,11 +196,12 @@ class FullyShardedDataParallel(torch.nn.Module):
     def _pre_forward(self, module, inputs):
         u = self._by_module[module]
         u.materialize()
-        # prefetch the NEXT unit's allgather so it overlaps this unit's
-        # forward compute (unit order = construction order)
+        # prefetch the next `prefetch` units' allgathers so they overlap
+        # this unit's forward compute (unit order = construction order)
         i = self._units.index(u)
-        if i + 1 < len(self._units):
-            self._units[i + 1].start_materialize()
+        for d in range(1, self.prefetch + 1):
+            if i + d < len(self._units):
+                self._units[i + d].start_materialize()
 
     def _post_forward(self, module, inputs, output):
         u = self._by_module[module]
@@ -206,10 +212,11 @@ class FullyShardedDataParallel(torch.nn.Module):
     def _pre_backward(self, module, grad_output):
         u = self._by_module[module]
         u.materialize()
-        # backward visits units in reverse: prefetch the PREVIOUS unit
+        # backward visits units in reverse: prefetch the PREVIOUS units
         i = self._units.index(u)
-        if i > 0:
-            self._units[i - 1].start_materialize()
+        for d in range(1, self.prefetch + 1):
+            if i - d >= 0:
+                self._units[i - d].start_materialize()
 
     def _grad_ready(self, p):
         if not self._sync_enabled:

@@ -143,3 +143,37 @@ def test_timeout_ws2():
         run_spmd(2, _timeout_worker)
     finally:
         os.environ.pop("MPI4TORCH_AMD_TIMEOUT_S", None)
+
+
+def _all_modes_worker(rank, world):
+    # every debug/lowering mode stacked: desync detector + p2p handshake
+    # + forced 4-phase pipelining + forced hierarchical lowering — the
+    # interaction matrix (detector host traffic interleaves phased gloo
+    # block traffic and handshake metadata)
+    os.environ["MPI4TORCH_AMD_DEBUG"] = "1"
+    os.environ["MPI4TORCH_AMD_PIPELINE_MB"] = "0.0001"
+    os.environ["MPI4TORCH_AMD_FORCE_HIERARCHICAL"] = "1"
+    import mpi4torch_amd as m
+
+    comm = m.COMM_WORLD
+    x = torch.full((3, rank + 1, 2), float(rank), dtype=torch.float64)
+    g = comm.Gather(x, 1, 0)
+    back = comm.Scatter(g, 1, rank + 1, 0)
+    assert (back == x).all()
+    comm.Allgather(x, 1)
+    ti = torch.arange(17, dtype=torch.int64) * (rank + 3)
+    comm.Allreduce(ti, m.MPI_BXOR)
+    t8 = (torch.randn(33) * 0.2).to(torch.float8_e4m3fn)
+    comm.Allreduce(t8, m.MPI_SUM)
+    pairs = torch.stack([torch.randn(9), torch.full((9,), float(rank))], -1)
+    comm.Allreduce(pairs, m.MPI_MAXLOC)
+    prev = (rank - 1 + world) % world
+    h = comm.Isend(x, (rank + 1) % world, 3)
+    buf = torch.empty(3, prev + 1, 2, dtype=torch.float64)
+    got = comm.Recv(m.JoinDummies(buf, [h.dummy]), prev, 3)
+    comm.Wait(m.JoinDummiesHandle(h, [got]))
+    assert (got == prev).all()
+
+
+def test_all_modes_stacked_ws5():
+    run_spmd(5, _all_modes_worker)

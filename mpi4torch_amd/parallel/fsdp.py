@@ -235,6 +235,25 @@ class FullyShardedDataParallel(torch.nn.Module):
 
     def finish_backward(self):
         """Wait all gradient reduce-scatters (call after loss.backward())."""
+        # a unit that only partially produced gradients never launched its
+        # reduce-scatter while peers may have — fail loudly instead of
+        # silently diverging (FSDP requires every unit parameter to get a
+        # gradient each backward; for data-dependent control flow use
+        # DistributedDataParallel(find_unused_parameters=True))
+        if self._sync_enabled:
+            fired = any(u.grad_handle is not None or
+                        (self.comm.size == 1 and u.shard.grad is not None)
+                        for u in self._units)
+            missing = sum(u.pending for u in self._units
+                          if 0 < u.pending <= len(u.params))
+            if fired and missing:
+                raise RuntimeError(
+                    "mpi4torch_amd FSDP: "
+                    f"{missing} parameter(s) received no gradient this "
+                    "backward while other units already launched their "
+                    "reduce-scatter — replicas would diverge. Every unit "
+                    "parameter must get a gradient each backward."
+                )
         for u in self._units:
             u.finish_grad_reduce(self.average)
 

@@ -128,6 +128,25 @@ class ShardedDataParallel(torch.nn.Module):
         """Wait bucket reductions, run the sharded optimizer, allgather the
         updated shards back into the module parameters."""
         P = self.comm.size
+        # a bucket that never filled would silently skip its reduce-scatter
+        # while peers (whose graphs used those params) issue theirs —
+        # divergence or a hang, so fail loudly. ZeRO-2 has no
+        # find_unused_parameters mode; use DistributedDataParallel for
+        # models with data-dependent control flow.
+        fired = [b for b in self._buckets
+                 if b.handle is not None or (P == 1 and b.shard.grad is not None)]
+        if fired and self._sync_enabled and any(
+                b.pending > 0 for b in self._buckets):
+            n_miss = sum(b.pending for b in self._buckets)
+            raise RuntimeError(
+                "mpi4torch_amd ShardedDataParallel: "
+                f"{n_miss} parameter(s) received no gradient this backward "
+                "while other buckets already launched their reduce-scatter "
+                "— replicas would diverge. Every parameter must get a "
+                "gradient each backward under ZeRO-2; for data-dependent "
+                "control flow use DistributedDataParallel("
+                "find_unused_parameters=True)."
+            )
         for b in self._buckets:
             if b.handle is not None:
                 g = self.comm.Wait(b.handle)

@@ -511,6 +511,49 @@ def test_fsdp_accumulation_ws2():
     run_spmd(2, _fsdp_accum_worker)
 
 
+def _zero2_unused_raises_worker(rank, world):
+    from mpi4torch_amd.parallel import ShardedDataParallel
+
+    torch.manual_seed(4)
+    net = _BranchyNet()
+    model = ShardedDataParallel(net, torch.optim.SGD, bucket_cap_mb=0,
+                                lr=0.1)
+    x = torch.randn(4, 8)
+    model(x, use_extra=False).sum().backward()
+    try:
+        model.step()
+        raise AssertionError("expected unused-parameter error")
+    except RuntimeError as e:
+        assert "no gradient" in str(e), e
+
+
+def test_zero2_unused_raises_ws3():
+    run_spmd(3, _zero2_unused_raises_worker)
+
+
+def _fsdp_unused_raises_worker(rank, world):
+    from mpi4torch_amd.parallel import FullyShardedDataParallel
+
+    torch.manual_seed(4)
+    net = _BranchyNet()
+    model = FullyShardedDataParallel(net, units=[net.trunk, net.extra])
+    x = torch.randn(4, 8)
+    out = model(x, use_extra=True)  # materializes both units' forwards
+    # build a loss that only uses the trunk's contribution gradient-wise
+    loss = net.trunk(x).sum()
+    loss.backward()
+    try:
+        model.finish_backward()
+        raise AssertionError("expected unused-parameter error")
+    except RuntimeError as e:
+        assert "no gradient" in str(e), e
+    del out
+
+
+def test_fsdp_unused_raises_ws3():
+    run_spmd(3, _fsdp_unused_raises_worker)
+
+
 def _fsdp_tied_worker(rank, world):
     import mpi4torch_amd as m  # noqa: F401
     from mpi4torch_amd.parallel import FullyShardedDataParallel

@@ -323,6 +323,32 @@ def section_comm_split(m, comm, device):
     assert (s == sum(members)).all()
 
 
+def section_hipgraph_multirank(m, comm, device):
+    # OPT-IN (M4A_TEST_HIPGRAPH=1): multi-rank hipGraph capture+replay of
+    # blocking collectives. Every rank captures the identical sequence
+    # (the ProcessGroupNCCL pattern); kept out of the default battery
+    # until validated on hardware (TODO.md item 3).
+    if os.environ.get("M4A_TEST_HIPGRAPH") != "1" or device.type != "cuda":
+        return
+    rank, world = comm.rank, comm.size
+    static_in = torch.rand(1 << 16, device=device)
+    x0 = static_in.clone()
+    s = torch.cuda.Stream()
+    s.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(s):
+        for _ in range(3):
+            comm.Allreduce(static_in, m.MPI_SUM)
+    torch.cuda.current_stream().wait_stream(s)
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        static_out = comm.Allreduce(static_in, m.MPI_SUM)
+    for i in range(3):
+        static_in.copy_(x0 * (i + 1))
+        g.replay()
+        torch.cuda.synchronize()
+        assert torch.allclose(static_out, x0 * (i + 1) * world)
+
+
 def section_ddp(m, comm, device):
     from mpi4torch_amd.parallel import DistributedDataParallel
 
@@ -393,6 +419,7 @@ def main():
     section_alltoall_pairwise(m, comm, device)
     log(rank, "comm_split subgroups")
     section_comm_split(m, comm, device)
+    section_hipgraph_multirank(m, comm, device)  # opt-in, see TODO.md
     log(rank, "DDP end-to-end")
     section_ddp(m, comm, device)
 

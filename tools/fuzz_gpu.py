@@ -29,7 +29,7 @@ import mpi4torch_amd as m
 m.init()
 comm = m.COMM_WORLD
 secs = float(sys.argv[1]) if len(sys.argv) > 1 else 180.0
-rng = random.Random(20260914)
+rng = random.Random(int(sys.argv[2]) if len(sys.argv) > 2 else 20260914)
 
 DTYPES = [torch.float32, torch.float64, torch.bfloat16, torch.float16,
           torch.int32, torch.int64, torch.uint8]

@@ -413,35 +413,45 @@ struct FMax {
   __device__ static float apply(float a, float b) { return fmaxf(a, b); }
 };
 
-// 8 elements per thread via 8-byte loads; fp32 accumulators.
+// 16 elements per thread via 16-byte (dwordx4) loads; fp32 accumulators.
+// (An 8-byte variant measured 5.0-5.1 TB/s; the wider granule lifts the
+// loads to the same dwordx4 form the 6.1 TB/s bitwise kernel uses.)
 template <typename Op, bool E5M2>
 __global__ __launch_bounds__(256) void fp8_reduce_kernel(
     const unsigned char* __restrict__ in, unsigned char* __restrict__ out,
     long long n, long long nranks) {
-  const long long nvec = n / 8;
+  const long long nvec = n / 16;
   const long long stride = (long long)gridDim.x * blockDim.x;
   for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
        i < nvec; i += stride) {
-    const unsigned long long* base =
-        reinterpret_cast<const unsigned long long*>(in);
-    unsigned long long w = __builtin_nontemporal_load(base + i);
-    float acc[8];
-    for (int k = 0; k < 8; ++k) {
-      acc[k] = fp8_to_f32<E5M2>((unsigned char)(w >> (8 * k)));
-    }
-    for (long long r = 1; r < nranks; ++r) {
-      unsigned long long wr =
-          __builtin_nontemporal_load(base + r * nvec + i);  // n % 8 == 0
-      for (int k = 0; k < 8; ++k) {
-        acc[k] = Op::apply(acc[k], fp8_to_f32<E5M2>((unsigned char)(wr >> (8 * k))));
+    const v4u* base = reinterpret_cast<const v4u*>(in);
+    v4u w = __builtin_nontemporal_load(base + i);
+    float acc[16];
+    for (int d = 0; d < 4; ++d) {
+      for (int k = 0; k < 4; ++k) {
+        acc[4 * d + k] =
+            fp8_to_f32<E5M2>((unsigned char)(w[d] >> (8 * k)));
       }
     }
-    unsigned long long o = 0;
-    for (int k = 0; k < 8; ++k) {
-      o |= (unsigned long long)f32_to_fp8<E5M2>(acc[k]) << (8 * k);
+    for (long long r = 1; r < nranks; ++r) {
+      v4u wr = __builtin_nontemporal_load(base + r * nvec + i);
+      for (int d = 0; d < 4; ++d) {
+        for (int k = 0; k < 4; ++k) {
+          acc[4 * d + k] = Op::apply(
+              acc[4 * d + k],
+              fp8_to_f32<E5M2>((unsigned char)(wr[d] >> (8 * k))));
+        }
+      }
     }
-    __builtin_nontemporal_store(
-        o, reinterpret_cast<unsigned long long*>(out) + i);
+    v4u o;
+    for (int d = 0; d < 4; ++d) {
+      unsigned int od = 0;
+      for (int k = 0; k < 4; ++k) {
+        od |= (unsigned int)f32_to_fp8<E5M2>(acc[4 * d + k]) << (8 * k);
+      }
+      o[d] = od;
+    }
+    __builtin_nontemporal_store(o, reinterpret_cast<v4u*>(out) + i);
   }
 }
 
@@ -464,10 +474,10 @@ __global__ __launch_bounds__(256) void fp8_reduce_kernel_scalar(
 template <typename Op, bool E5M2>
 void launch_fp8_reduce_t(const void* in, void* out, int64_t n, int nranks,
                          hipStream_t stream) {
-  const bool vec = (n % 8 == 0) &&
-                   (reinterpret_cast<uintptr_t>(in) % 8 == 0) &&
-                   (reinterpret_cast<uintptr_t>(out) % 8 == 0);
-  const long long work = vec ? n / 8 : n;
+  const bool vec = (n % 16 == 0) &&
+                   (reinterpret_cast<uintptr_t>(in) % 16 == 0) &&
+                   (reinterpret_cast<uintptr_t>(out) % 16 == 0);
+  const long long work = vec ? n / 16 : n;
   long long blocks = (work + 255) / 256;
   if (blocks > 4096) blocks = 4096;
   if (blocks < 1) blocks = 1;

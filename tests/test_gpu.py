@@ -178,7 +178,7 @@ def test_fp8_reduce_kernel(world1):
     m = world1
     # fused fp8 reduce (fp32 accumulation) vs plain PyTorch fp32 reference
     for enc in (torch.float8_e4m3fn, torch.float8_e5m2):
-        for n in (4096, 4099):  # vector path and scalar-tail path
+        for n in (4096, 4104, 4099):  # 16B-vector, 8-not-16, scalar paths
             stacked = (torch.randn(5, n, device="cuda") * 0.25).to(enc)
             ref_in = stacked.float()
             cases = [

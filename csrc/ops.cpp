@@ -377,10 +377,10 @@ bool native_reduce_dtype(const Transport& tr, at::ScalarType t) {
         return true;
       case at::kFloat8_e4m3fn:
       case at::kFloat8_e5m2: {
-        // Default: the fused allgather + fp32-accumulating CDNA4 kernel
-        // (one quantization; equal wire bytes to an fp32-upcast allreduce
-        // at P=8). Opt into RCCL-native fp8 rings with
-        // MPI4TORCH_AMD_NATIVE_FP8=1 (runtime-probed).
+        // Default: the hierarchical block exchange + fp32-accumulating
+        // CDNA4 kernel (one quantization, ~numel staging). Opt into
+        // RCCL-native fp8 rings with MPI4TORCH_AMD_NATIVE_FP8=1
+        // (runtime-probed).
         static const bool want_native = []() {
           const char* e = std::getenv("MPI4TORCH_AMD_NATIVE_FP8");
           return e && e[0] == '1';

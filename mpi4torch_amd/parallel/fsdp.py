@@ -264,6 +264,24 @@ class FullyShardedDataParallel(torch.nn.Module):
         for u in self._units:
             u.shard.grad = None
 
+    def sharded_state_dict(self):
+        """This rank's parameter shards (one per unit). The optimizer the
+        caller runs over shard_parameters() carries its own state_dict.
+        Pair with utils.checkpoint.save_sharded_checkpoint; reload
+        requires the SAME world size."""
+        return {"units": [u.shard.detach().cpu() for u in self._units]}
+
+    @torch.no_grad()
+    def load_sharded_state_dict(self, state):
+        for u, s in zip(self._units, state["units"]):
+            if u.prefetch_handle is not None:
+                self.comm.Wait(u.prefetch_handle)  # drain stale prefetch
+                u.prefetch_handle = None
+            u.shard.copy_(s.to(u.shard.device))
+            # drop any materialized full buffer so the next use
+            # re-allgathers from the restored shards
+            u.free()
+
     def no_sync(self):
         """Gradient-accumulation context: local .grad accumulates; the
         final backward outside the context reduces the sums. (Parameters

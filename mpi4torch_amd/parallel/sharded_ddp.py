@@ -167,6 +167,33 @@ class ShardedDataParallel(torch.nn.Module):
                 off += n
         self._reset_pending()
 
+    def sharded_state_dict(self):
+        """This rank's shard of the training state (bucket parameter
+        shards + the sharded inner optimizer). Pair with
+        utils.checkpoint.save_sharded_checkpoint; reload requires the
+        SAME world size."""
+        return {
+            "buckets": [b.shard.detach().cpu() for b in self._buckets],
+            "inner": self.optimizer.state_dict(),
+        }
+
+    @torch.no_grad()
+    def load_sharded_state_dict(self, state):
+        for b, s in zip(self._buckets, state["buckets"]):
+            b.shard.copy_(s.to(b.shard.device))
+        self.optimizer.load_state_dict(state["inner"])
+        # rebroadcast full parameters from the restored shards
+        P = self.comm.size
+        for b in self._buckets:
+            local = b.shard.detach()
+            if local.dtype != b.params[0].dtype:
+                local = local.to(b.params[0].dtype)
+            full = self.comm.Allgather(local, 0) if P > 1 else local
+            off = 0
+            for q, n in zip(b.params, b.numels):
+                q.data.copy_(full[off:off + n].view_as(q))
+                off += n
+
     def no_sync(self):
         """Context manager for gradient-accumulation microbatches: local
         gradients accumulate in .grad; the final backward OUTSIDE the

@@ -1,6 +1,7 @@
 from .timing import CollectiveTimer, algbw_gbps, busbw_gbps
 from .tracing import trace, TracedCommunicator, TraceRecord
-from .checkpoint import save_checkpoint, load_checkpoint
+from .checkpoint import (save_checkpoint, load_checkpoint,
+                         save_sharded_checkpoint, load_sharded_checkpoint)
 from .clip import clip_grad_norm_sharded
 
 __all__ = [
@@ -12,5 +13,7 @@ __all__ = [
     "TraceRecord",
     "save_checkpoint",
     "load_checkpoint",
+    "save_sharded_checkpoint",
+    "load_sharded_checkpoint",
     "clip_grad_norm_sharded",
 ]

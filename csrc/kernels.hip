@@ -2,9 +2,13 @@
 //
 // These kernels are the MI355X-native replacement for the CPU-side MPI
 // derived-datatype marshaling the reference uses for axis-aware collectives
-// (reference csrc/extension.cpp:556-577, 691-712, 839-861): strided slab
-// pack/unpack between tensors and contiguous RCCL staging buffers, plus a
-// local bitwise reduction for the MPI_BAND/BOR/BXOR ops RCCL lacks.
+// (reference csrc/extension.cpp:556-577, 691-712, 839-861) and for the
+// reduction ops RCCL lacks. Four families: strided slab pack/unpack
+// between tensors and contiguous RCCL staging buffers; bitwise reduction
+// (MPI_BAND/BOR/BXOR); fused fp8 reduction with fp32 accumulation and a
+// single quantization; and the MINLOC/MAXLOC pair arg-reduce. The latter
+// three are the local tails of the hierarchical allreduce lowering
+// (csrc/ops.cpp hierarchical_allreduce).
 //
 // Written directly for gfx950: 64-wide wavefronts, 256-thread workgroups,
 // 16-byte (dwordx4) vectorized global accesses, grids sized >> 256

@@ -469,8 +469,6 @@ def test_ireducescatter_gpu(world1):
 def test_zero_family_gpu(world1):
     # ZeRO-1/2/3 on CUDA at world 1 must match a plain local Adam exactly
     # (exercises CUDA storage-resize, Ireducescatter and allgather paths)
-    import copy
-
     from mpi4torch_amd.parallel import (FullyShardedDataParallel,
                                         ShardedDataParallel,
                                         ZeroRedundancyOptimizer)

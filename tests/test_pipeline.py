@@ -34,8 +34,6 @@ def _pipeline_worker(rank, world):
     )
 
     # dense reference: same segments chained on one process
-    import copy
-
     torch.manual_seed(31)
     ref_segments = [
         torch.nn.Sequential(torch.nn.Linear(d, d, dtype=torch.double),

@@ -8,7 +8,6 @@ from spmd import run_spmd
 
 
 def _tp_mlp_worker(rank, world):
-    import mpi4torch_amd as m
     from mpi4torch_amd.parallel.tp import TensorParallelMLP
 
     torch.manual_seed(7)  # identical on all ranks

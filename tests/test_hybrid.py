@@ -1,0 +1,91 @@
+"""Hybrid parallelism: TP x DP over comm_split sub-communicators.
+
+The production topology no single-communicator test covers: world 4
+splits into two tensor-parallel groups ({0,1}, {2,3} — one model replica
+each) and two data-parallel groups ({0,2}, {1,3} — pairing corresponding
+TP shards). A TensorParallelMLP runs over the TP communicator, wrapped
+in DistributedDataParallel over the DP communicator, so BOTH
+communicators issue collectives in the same forward/backward. Gradients
+must equal the dense single-process reference averaged over the DP data
+shards, sliced per TP rank (fp64, 1e-10).
+"""
+
+import torch
+
+from spmd import run_spmd
+
+
+def _hybrid_worker(rank, world, tp_size, h):
+    import mpi4torch_amd as m
+    from mpi4torch_amd.parallel import DistributedDataParallel
+    from mpi4torch_amd.parallel.tp import TensorParallelMLP
+
+    comm = m.COMM_WORLD
+    dp_size = world // tp_size
+    tp_color = rank // tp_size   # consecutive ranks form one replica
+    dp_color = rank % tp_size    # same TP position across replicas
+    tp_comm = m.comm_split(comm, tp_color)
+    dp_comm = m.comm_split(comm, dp_color)
+    assert tp_comm.size == tp_size and dp_comm.size == dp_size
+
+    torch.manual_seed(7)
+    d, b = 6, 4
+    w1 = torch.randn(h, d, dtype=torch.double)
+    b1 = torch.randn(h, dtype=torch.double)
+    w2 = torch.randn(d, h, dtype=torch.double)
+
+    mlp = TensorParallelMLP(d, h, comm=tp_comm,
+                            activation=torch.relu).double()
+    mlp.up.shard_from_full(w1, b1)
+    mlp.down.shard_from_full(w2)
+    with torch.no_grad():
+        mlp.down.bias.zero_()
+    model = DistributedDataParallel(mlp, comm=dp_comm, bucket_cap_mb=0)
+
+    # data shard per DP replica (= per TP group); identical inside one
+    # TP group, different across replicas
+    torch.manual_seed(1000 + tp_color)
+    x = torch.randn(b, d, dtype=torch.double)
+
+    loss = (model(x) ** 2).sum()
+    loss.backward()
+    model.finish_gradient_sync()
+
+    # dense reference: per-shard gradients, then the DP average
+    grads = []
+    for s in range(dp_size):
+        torch.manual_seed(1000 + s)
+        xs = torch.randn(b, d, dtype=torch.double)
+        w1r = w1.clone().requires_grad_()
+        b1r = b1.clone().requires_grad_()
+        w2r = w2.clone().requires_grad_()
+        yr = torch.relu(xs @ w1r.t() + b1r) @ w2r.t()
+        (yr ** 2).sum().backward()
+        grads.append((w1r.grad, b1r.grad, w2r.grad))
+    g1 = sum(g[0] for g in grads) / dp_size
+    gb = sum(g[1] for g in grads) / dp_size
+    g2 = sum(g[2] for g in grads) / dp_size
+
+    tpr = tp_comm.rank
+    lo, hi = tpr * mlp.up.out_local, (tpr + 1) * mlp.up.out_local
+    assert torch.allclose(mlp.up.linear.weight.grad, g1[lo:hi], atol=1e-10)
+    assert torch.allclose(mlp.up.linear.bias.grad, gb[lo:hi], atol=1e-10)
+    lo2, hi2 = tpr * mlp.down.in_local, (tpr + 1) * mlp.down.in_local
+    assert torch.allclose(mlp.down.linear.weight.grad, g2[:, lo2:hi2],
+                          atol=1e-10)
+
+    # one optimizer step: all four ranks' shards stay consistent with the
+    # dense trajectory (TP shards per group, identical across DP pairs)
+    opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    opt.step()
+    got = dp_comm.Bcast_(mlp.up.linear.weight.data.clone(), 0)
+    assert torch.allclose(mlp.up.linear.weight.data, got, atol=1e-12), \
+        "DP pair shards diverged after step"
+
+
+def test_hybrid_tp2_dp2_ws4():
+    run_spmd(4, _hybrid_worker, 2, 8)
+
+
+def test_hybrid_tp3_dp2_ws6():
+    run_spmd(6, _hybrid_worker, 3, 9)

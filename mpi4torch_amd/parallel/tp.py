@@ -103,7 +103,11 @@ class RowParallelLinear(torch.nn.Module):
         self.linear = torch.nn.Linear(self.in_local, out_features, bias=False)
         self.bias = (torch.nn.Parameter(torch.zeros(out_features))
                      if bias else None)
-        _register_tp_grad_scale(self, world)
+        # the /P correction applies only BELOW the output Allreduce (whose
+        # adjoint multiplies upstream gradients by P). The bias adds AFTER
+        # it: its gradient is already at true scale on every rank, so
+        # scaling it would under-train the bias by the TP degree.
+        _register_tp_grad_scale(self.linear, world)
 
     def forward(self, x_local: torch.Tensor) -> torch.Tensor:
         partial = self.linear(x_local)

@@ -83,6 +83,86 @@ def _hybrid_worker(rank, world, tp_size, h):
         "DP pair shards diverged after step"
 
 
+def _hybrid_fsdp_worker(rank, world):
+    # FSDP over the DP communicator with TP modules (over the TP
+    # communicator) inside its units: parameter shards at rest ride dp,
+    # forward/backward collectives ride tp, gradient reduce-scatters ride
+    # dp — three communicators live at once (world + tp + dp)
+    import mpi4torch_amd as m
+    from mpi4torch_amd.parallel import FullyShardedDataParallel
+    from mpi4torch_amd.parallel.tp import TensorParallelMLP
+
+    comm = m.COMM_WORLD
+    tp_size, dp_size = 2, world // 2
+    tp_color, dp_color = rank // tp_size, rank % tp_size
+    tp_comm = m.comm_split(comm, tp_color)
+    dp_comm = m.comm_split(comm, dp_color)
+
+    torch.manual_seed(7)
+    d, h, b = 6, 8, 4
+    w1 = torch.randn(h, d, dtype=torch.double)
+    b1 = torch.randn(h, dtype=torch.double)
+    w2 = torch.randn(d, h, dtype=torch.double)
+
+    mlp = TensorParallelMLP(d, h, comm=tp_comm,
+                            activation=torch.relu).double()
+    mlp.up.shard_from_full(w1, b1)
+    mlp.down.shard_from_full(w2)
+    with torch.no_grad():
+        mlp.down.bias.zero_()
+    model = FullyShardedDataParallel(mlp, units=[mlp.up, mlp.down],
+                                     comm=dp_comm)
+
+    torch.manual_seed(1000 + tp_color)
+    x = torch.randn(b, d, dtype=torch.double)
+    loss = (model(x) ** 2).sum()
+    loss.backward()
+    model.finish_backward()
+
+    # dense reference per DP shard, averaged (incl. the Row bias, which
+    # is full-size and replicated across TP)
+    grads = []
+    for s in range(dp_size):
+        torch.manual_seed(1000 + s)
+        xs = torch.randn(b, d, dtype=torch.double)
+        w1r = w1.clone().requires_grad_()
+        b1r = b1.clone().requires_grad_()
+        w2r = w2.clone().requires_grad_()
+        b2r = torch.zeros(d, dtype=torch.double, requires_grad=True)
+        ((torch.relu(xs @ w1r.t() + b1r) @ w2r.t() + b2r) ** 2
+         ).sum().backward()
+        grads.append((w1r.grad, b1r.grad, w2r.grad, b2r.grad))
+    g1 = sum(g[0] for g in grads) / dp_size
+    gb = sum(g[1] for g in grads) / dp_size
+    g2 = sum(g[2] for g in grads) / dp_size
+    gb2 = sum(g[3] for g in grads) / dp_size
+
+    tpr = tp_comm.rank
+    lo, hi = tpr * mlp.up.out_local, (tpr + 1) * mlp.up.out_local
+    lo2, hi2 = tpr * mlp.down.in_local, (tpr + 1) * mlp.down.in_local
+    # per-parameter dense expectations, keyed by identity (the unit's flat
+    # layout follows module.parameters() order — direct Parameters come
+    # BEFORE child modules', e.g. RowParallel's bias precedes its weight)
+    want_by_param = {
+        mlp.up.linear.weight: g1[lo:hi],
+        mlp.up.linear.bias: gb[lo:hi],
+        mlp.down.linear.weight: g2[:, lo2:hi2],
+        mlp.down.bias: gb2,
+    }
+    for u in model._units:
+        full = dp_comm.Allgather(u.shard.grad.detach(), 0)
+        off = 0
+        for p, n in zip(u.params, u.numels):
+            got = full[off:off + n].view_as(p)
+            want = want_by_param[p]
+            assert torch.allclose(got, want, atol=1e-10), (p.shape, off)
+            off += n
+
+
+def test_hybrid_fsdp_tp_ws4():
+    run_spmd(4, _hybrid_fsdp_worker)
+
+
 def test_hybrid_tp2_dp2_ws4():
     run_spmd(4, _hybrid_worker, 2, 8)
 

@@ -28,13 +28,18 @@ def _tp_mlp_worker(rank, world):
     loss = (y ** 2).sum()
     loss.backward()
 
-    # dense reference
+    # dense reference (incl. the Row bias, replicated across TP ranks —
+    # it adds AFTER the output allreduce so its grad must be at TRUE
+    # dense scale, not /P; regression test for the scaling-hook bug the
+    # FSDPxTP hybrid test caught)
     x_ref = x.detach().clone().requires_grad_()
     w1_ref = w1.clone().requires_grad_()
     b1_ref = b1.clone().requires_grad_()
     w2_ref = w2.clone().requires_grad_()
-    y_ref = torch.relu(x_ref @ w1_ref.t() + b1_ref) @ w2_ref.t()
+    b2_ref = torch.zeros(d_model, dtype=torch.double, requires_grad=True)
+    y_ref = torch.relu(x_ref @ w1_ref.t() + b1_ref) @ w2_ref.t() + b2_ref
     ((y_ref ** 2).sum()).backward()
+    assert torch.allclose(mlp.down.bias.grad, b2_ref.grad, atol=1e-10)
 
     assert torch.allclose(y, y_ref, atol=1e-10), (y - y_ref).abs().max()
     # input gradient: exact (entry collective sums shard contributions)

@@ -163,6 +163,81 @@ def test_hybrid_fsdp_tp_ws4():
     run_spmd(4, _hybrid_fsdp_worker)
 
 
+def _hybrid_moe_dp_worker(rank, world):
+    # EP x DP: expert-parallel MoE over the EP communicator inside DDP
+    # over the DP communicator — the production MoE layout. Experts that
+    # receive no tokens on a replica produce no gradients, so this is
+    # also the real-world consumer of find_unused_parameters=True.
+    import mpi4torch_amd as m
+    from mpi4torch_amd.models.moe import ExpertParallelMoE
+    from mpi4torch_amd.parallel import DistributedDataParallel
+
+    comm = m.COMM_WORLD
+    ep_size = 2
+    ep_color, dp_color = rank // ep_size, rank % ep_size
+    ep_comm = m.comm_split(comm, ep_color)
+    dp_comm = m.comm_split(comm, dp_color)
+    dp_size = dp_comm.size
+
+    torch.manual_seed(3)
+    d_model, n_experts, N = 8, 2 * ep_size, 12
+    moe = ExpertParallelMoE(d_model, n_experts, d_hidden=16,
+                            comm=ep_comm).double()
+    model = DistributedDataParallel(moe, comm=dp_comm, bucket_cap_mb=0,
+                                    find_unused_parameters=True)
+
+    torch.manual_seed(50 + ep_color * 100 + ep_comm.rank)
+    x = torch.randn(N, d_model, dtype=torch.double)
+    y = model(x)
+    y.square().sum().backward()
+
+    # per-replica dense forward reference (same construction as the
+    # single-comm MoE parity test)
+    torch.manual_seed(3)
+    full = ExpertParallelMoE(d_model, n_experts, d_hidden=16,
+                             comm=ep_comm).double()
+    logits = full.router(x)
+    gates = torch.softmax(logits, dim=-1)
+    expert = torch.argmax(gates, dim=-1)
+    gate = gates.gather(1, expert.unsqueeze(1)).squeeze(1)
+    y_ref = torch.zeros_like(x)
+    for gid in range(n_experts):
+        mask = expert == gid
+        if bool(mask.any()):
+            idx = mask.nonzero(as_tuple=True)[0]
+            mod = full.experts[gid % full.experts_per_rank]
+            y_ref = y_ref.index_copy(0, idx,
+                                     mod(x.index_select(0, idx)))
+    y_ref = y_ref * gate.unsqueeze(1)
+    assert torch.allclose(y.detach(), y_ref, atol=1e-9)
+
+    model.finish_gradient_sync()
+
+    # DP-paired ranks (same EP position, different replicas) must agree
+    # on every present gradient after the find_unused sync
+    for p in moe.parameters():
+        if p.grad is None:
+            presence = dp_comm.Allreduce(
+                torch.zeros(1, dtype=torch.double), m.MPI_MAX)
+            assert (presence == 0).all(), "grad present on peer but None here"
+            continue
+        presence = dp_comm.Allreduce(
+            torch.ones(1, dtype=torch.double), m.MPI_MAX)
+        g_ref = dp_comm.Bcast_(p.grad.clone(), 0)
+        assert torch.allclose(p.grad, g_ref, atol=1e-9), "dp grads diverged"
+
+    # one step: replicas stay consistent
+    opt = torch.optim.SGD([p for p in moe.parameters()], lr=0.01)
+    opt.step()
+    for p in moe.parameters():
+        ref = dp_comm.Bcast_(p.data.clone(), 0)
+        assert torch.allclose(p.data, ref, atol=1e-12)
+
+
+def test_hybrid_moe_dp_ws4():
+    run_spmd(4, _hybrid_moe_dp_worker)
+
+
 def test_hybrid_tp2_dp2_ws4():
     run_spmd(4, _hybrid_worker, 2, 8)
 

@@ -349,6 +349,64 @@ def section_hipgraph_multirank(m, comm, device):
         assert torch.allclose(static_out, x0 * (i + 1) * world)
 
 
+def section_hybrid_tp_dp(m, comm, device):
+    # TP x DP over comm_split: TensorParallelMLP (tp) inside DDP (dp),
+    # gradients vs the dense reference averaged over DP shards
+    from mpi4torch_amd.parallel import DistributedDataParallel
+    from mpi4torch_amd.parallel.tp import TensorParallelMLP
+
+    rank, world = comm.rank, comm.size
+    if world < 4 or world % 2 != 0:
+        return
+    tp_size = 2
+    dp_size = world // tp_size
+    tp_color, dp_color = rank // tp_size, rank % tp_size
+    tp_comm = m.comm_split(comm, tp_color)
+    dp_comm = m.comm_split(comm, dp_color)
+
+    torch.manual_seed(7)
+    d, h, b = 16, 32, 8
+    w1 = torch.randn(h, d, dtype=torch.double, device=device)
+    b1 = torch.randn(h, dtype=torch.double, device=device)
+    w2 = torch.randn(d, h, dtype=torch.double, device=device)
+    mlp = TensorParallelMLP(d, h, comm=tp_comm,
+                            activation=torch.relu).double().to(device)
+    mlp.up.shard_from_full(w1, b1)
+    mlp.down.shard_from_full(w2)
+    with torch.no_grad():
+        mlp.down.bias.zero_()
+    model = DistributedDataParallel(mlp, comm=dp_comm, bucket_cap_mb=0)
+    torch.manual_seed(1000 + tp_color)
+    x = torch.randn(b, d, dtype=torch.double, device=device)
+    (model(x) ** 2).sum().backward()
+    model.finish_gradient_sync()
+
+    grads = []
+    for s in range(dp_size):
+        torch.manual_seed(1000 + s)
+        xs = torch.randn(b, d, dtype=torch.double, device=device)
+        w1r = w1.clone().requires_grad_()
+        b1r = b1.clone().requires_grad_()
+        w2r = w2.clone().requires_grad_()
+        b2r = torch.zeros(d, dtype=torch.double, device=device,
+                          requires_grad=True)
+        ((torch.relu(xs @ w1r.t() + b1r) @ w2r.t() + b2r) ** 2
+         ).sum().backward()
+        grads.append((w1r.grad, b1r.grad, w2r.grad, b2r.grad))
+    g1 = sum(g[0] for g in grads) / dp_size
+    gb = sum(g[1] for g in grads) / dp_size
+    g2 = sum(g[2] for g in grads) / dp_size
+    gb2 = sum(g[3] for g in grads) / dp_size
+    tpr = tp_comm.rank
+    lo, hi = tpr * mlp.up.out_local, (tpr + 1) * mlp.up.out_local
+    lo2, hi2 = tpr * mlp.down.in_local, (tpr + 1) * mlp.down.in_local
+    assert torch.allclose(mlp.up.linear.weight.grad, g1[lo:hi], atol=1e-9)
+    assert torch.allclose(mlp.up.linear.bias.grad, gb[lo:hi], atol=1e-9)
+    assert torch.allclose(mlp.down.linear.weight.grad, g2[:, lo2:hi2],
+                          atol=1e-9)
+    assert torch.allclose(mlp.down.bias.grad, gb2, atol=1e-9)
+
+
 def section_ddp(m, comm, device):
     from mpi4torch_amd.parallel import DistributedDataParallel
 
@@ -419,6 +477,8 @@ def main():
     section_alltoall_pairwise(m, comm, device)
     log(rank, "comm_split subgroups")
     section_comm_split(m, comm, device)
+    log(rank, "hybrid TP x DP")
+    section_hybrid_tp_dp(m, comm, device)
     section_hipgraph_multirank(m, comm, device)  # opt-in, see TODO.md
     log(rank, "DDP end-to-end")
     section_ddp(m, comm, device)

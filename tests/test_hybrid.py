@@ -283,6 +283,81 @@ def test_hybrid_ulysses_dp_ws4():
     run_spmd(4, _hybrid_ulysses_dp_worker)
 
 
+def _hybrid_pp_dp_worker(rank, world):
+    # PP x DP: a 2-stage GPipe over the PP communicator, replicated twice;
+    # stage gradients average over the DP communicator after the drain.
+    # Pipeline p2p (fwd + adjoint channels) and DP collectives share the
+    # process; gradients must equal the dense 2-segment chain's, averaged
+    # over the DP data shards.
+    import mpi4torch_amd as m
+    from mpi4torch_amd.parallel import GPipe
+
+    comm = m.COMM_WORLD
+    pp_size = 2
+    pp_color, dp_color = rank // pp_size, rank % pp_size
+    pp_comm = m.comm_split(comm, pp_color)   # replicas {0,1}, {2,3}
+    dp_comm = m.comm_split(comm, dp_color)   # stage pairs {0,2}, {1,3}
+    dp_size = dp_comm.size
+
+    d, batch, n_mb = 5, 3, 4
+    torch.manual_seed(31)
+    segments = [
+        torch.nn.Sequential(torch.nn.Linear(d, d, dtype=torch.double),
+                            torch.nn.Tanh())
+        for _ in range(pp_size)
+    ]
+    stage = segments[pp_comm.rank]
+    pipe = GPipe(stage, recv_shape=(batch, d), recv_dtype=torch.double,
+                 comm=pp_comm)
+
+    def make_data(shard):
+        torch.manual_seed(99 + shard)
+        data = [torch.randn(batch, d, dtype=torch.double)
+                for _ in range(n_mb)]
+        tg = [torch.randn(batch, d, dtype=torch.double)
+              for _ in range(n_mb)]
+        return data, tg
+
+    data, targets = make_data(pp_color)
+
+    def loss_fn(y, i):
+        return ((y - targets[i]) ** 2).sum()
+
+    pipe.run(
+        microbatches=data if pipe.is_first else None,
+        loss_fn=loss_fn if pipe.is_last else None,
+        n_microbatches=n_mb,
+    )
+    # DP gradient sync of this stage's accumulated gradients
+    with torch.no_grad():
+        for p in stage.parameters():
+            p.grad = dp_comm.Allreduce(p.grad, m.MPI_SUM) / dp_size
+
+    # dense reference, averaged over the DP shards
+    torch.manual_seed(31)
+    ref_segments = [
+        torch.nn.Sequential(torch.nn.Linear(d, d, dtype=torch.double),
+                            torch.nn.Tanh())
+        for _ in range(pp_size)
+    ]
+    full = torch.nn.Sequential(*ref_segments)
+    for shard in range(dp_size):
+        sdata, stg = make_data(shard)
+        total = torch.zeros((), dtype=torch.double)
+        for i in range(n_mb):
+            total = total + ((full(sdata[i]) - stg[i]) ** 2).sum()
+        total.backward()
+    for p, q in zip(stage.parameters(),
+                    ref_segments[pp_comm.rank].parameters()):
+        want = q.grad / dp_size
+        assert torch.allclose(p.grad, want, atol=1e-9), (
+            rank, (p.grad - want).abs().max())
+
+
+def test_hybrid_pp_dp_ws4():
+    run_spmd(4, _hybrid_pp_dp_worker)
+
+
 def test_hybrid_tp2_dp2_ws4():
     run_spmd(4, _hybrid_worker, 2, 8)
 

@@ -257,11 +257,6 @@ def main():
         ctypes.CDLL(None).fflush(None)
     except Exception:
         pass
-    if rank == 0 and extra_records:
-        os.makedirs("gpurun_out", exist_ok=True)
-        with open("gpurun_out/bench_all.jsonl", "w") as f:
-            for r in extra_records:
-                f.write(json.dumps({**r, "n_gpus": world}) + "\n")
     if rank == 0:
         out = {
             "metric": result["metric"],
@@ -279,6 +274,15 @@ def main():
             "data": "synthetic",
             "config": result["config"],
         }
+        if extra_records:
+            # --all-configs: persist every config's record (headline
+            # included) for the profiles/ archive; stdout still carries
+            # exactly ONE JSON line (the driver's contract)
+            os.makedirs("gpurun_out", exist_ok=True)
+            with open("gpurun_out/bench_all.jsonl", "w") as f:
+                for r in extra_records:
+                    f.write(json.dumps({**r, "n_gpus": world}) + "\n")
+                f.write(json.dumps(out) + "\n")
         print(json.dumps(out), flush=True)
 
     import torch.distributed as dist

@@ -26,7 +26,7 @@ from typing import List, Optional
 
 import torch
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
 
 from . import _C  # registers torch.classes.mpi4torch_amd.* and torch.ops.mpi4torch_amd.*
 

@@ -98,18 +98,40 @@ def bench_allreduce(m4a, comm, device, args):
 
     elapsed = time_steps(step, args.steps, args.warmup, device)
     t_step = elapsed / args.steps
+
+    config = {
+        "model": "allreduce_1GiB_bf16_fwd_bwd",
+        "tensor_mib": args.size_mib,
+        "global_batch": None,
+        "seq_len": None,
+        "parallelism": f"dp{comm.size}",
+    }
+    if comm.size == 1 and device.type == "cuda":
+        # The timed region holds a REAL ncclAllReduce (force_full_path);
+        # at nranks=1 RCCL implements it as an internal copy whose speed
+        # varies wildly across boxes (profiles/rccl_1rank_box_variance.md:
+        # 1.15-9.2 ms/GiB measured for identical code). Record the
+        # framework's DEFAULT world-1 path alongside so the record is
+        # self-explanatory; the N>1 numbers are the real communication.
+        m4a = __import__("mpi4torch_amd")
+        m4a._C.force_full_path(False)
+        try:
+            fp_elapsed = time_steps(step, 5, 2, device)
+            config["n1_default_path_ms"] = round(fp_elapsed / 5 * 1e3, 3)
+            config["n1_note"] = (
+                "value times RCCL's nranks=1 internal copy "
+                "(box-variable; see profiles/rccl_1rank_box_variance.md); "
+                "n1_default_path_ms is the framework's default world-1 "
+                "path on the same box")
+        finally:
+            m4a._C.force_full_path(True)
+
     nbytes = n_elems * 2
     return {
         "metric": "allreduce_fwdbwd_algbw_GBps",
         "value": comm.size * 2 * nbytes / t_step / 1e9,
         "unit": "GB/s",
-        "config": {
-            "model": "allreduce_1GiB_bf16_fwd_bwd",
-            "tensor_mib": args.size_mib,
-            "global_batch": None,
-            "seq_len": None,
-            "parallelism": f"dp{comm.size}",
-        },
+        "config": config,
     }, t_step
 
 
